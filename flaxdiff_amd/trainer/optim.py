@@ -1,0 +1,154 @@
+"""Flat-buffer AdamW + EMA optimizer.
+
+All parameters are re-parented into ONE contiguous fp32 master buffer, laid
+out in REVERSE registration order so backward fills the matching flat
+gradient buffer front-to-back (enables bucketed all-reduce overlap —
+parallel.GradBucketSynchronizer). The Adam update + EMA lerp + grad-mean
+scaling run as ONE fused HIP kernel pass over the flat buffers on GPU
+(ops.fused_adamw_ema); on CPU the same math runs as flat torch ops.
+
+Replaces the reference's optax.adamw + apply_ema tree_map
+(diffusion_trainer.py:31-37, training.py:594-608).
+"""
+from __future__ import annotations
+
+import math
+from typing import Callable, Iterable, List, Optional
+
+import torch
+
+from .. import ops
+
+
+class FlatAdamWEMA:
+    def __init__(self, module: torch.nn.Module, lr: float = 2.7e-4,
+                 betas=(0.9, 0.999), eps: float = 1e-8, weight_decay: float = 0.0,
+                 ema_decay: float = 0.999, lr_schedule: Optional[Callable[[int], float]] = None,
+                 grad_clip_norm: Optional[float] = None):
+        self.lr = lr
+        self.beta1, self.beta2 = betas
+        self.eps = eps
+        self.weight_decay = weight_decay
+        self.ema_decay = ema_decay
+        self.lr_schedule = lr_schedule
+        self.grad_clip_norm = grad_clip_norm
+        self.step_count = 0
+
+        params = [p for p in module.parameters() if p.requires_grad]
+        params = list(reversed(params))  # backward-production order
+        total = sum(p.numel() for p in params)
+        device = params[0].device if params else torch.device("cpu")
+
+        self.flat = torch.zeros(total, dtype=torch.float32, device=device)
+        self.flat_grad = torch.zeros(total, dtype=torch.float32, device=device)
+        self.exp_avg = torch.zeros(total, dtype=torch.float32, device=device)
+        self.exp_avg_sq = torch.zeros(total, dtype=torch.float32, device=device)
+        self.ema = torch.zeros(total, dtype=torch.float32, device=device)
+
+        self.params: List[torch.nn.Parameter] = params
+        self.offsets: List[int] = []
+        off = 0
+        with torch.no_grad():
+            for p in params:
+                n = p.numel()
+                self.flat[off:off + n].copy_(p.data.reshape(-1).float())
+                p.data = self.flat[off:off + n].view(p.shape)
+                p.grad = self.flat_grad[off:off + n].view(p.shape)
+                self.offsets.append(off)
+                off += n
+        self.ema.copy_(self.flat)
+        self.total = total
+
+    # ------------------------------------------------------------------
+    def zero_grad(self):
+        self.flat_grad.zero_()
+
+    def current_lr(self) -> float:
+        if self.lr_schedule is not None:
+            return self.lr_schedule(self.step_count)
+        return self.lr
+
+    @torch.no_grad()
+    def step(self, grad_scale: float = 1.0):
+        """grad_scale multiplies gradients before use (e.g. 1/world_size so the
+        all-reduce SUM becomes the reference's pmean)."""
+        self.step_count += 1
+        lr = self.current_lr()
+
+        if self.grad_clip_norm is not None:
+            gnorm = self.flat_grad.norm() * grad_scale
+            clip = self.grad_clip_norm / (float(gnorm) + 1e-6)
+            if clip < 1.0:
+                grad_scale = grad_scale * clip
+
+        if self.flat.is_cuda and ops.hip_available():
+            ops.fused_adamw_ema(self.flat, self.flat_grad, self.exp_avg,
+                                self.exp_avg_sq, self.ema, None,
+                                lr=lr, beta1=self.beta1, beta2=self.beta2,
+                                eps=self.eps, weight_decay=self.weight_decay,
+                                step=self.step_count, ema_decay=self.ema_decay,
+                                grad_scale=grad_scale)
+            return
+
+        g = self.flat_grad
+        if grad_scale != 1.0:
+            g = g * grad_scale
+        self.exp_avg.mul_(self.beta1).add_(g, alpha=1 - self.beta1)
+        self.exp_avg_sq.mul_(self.beta2).addcmul_(g, g, value=1 - self.beta2)
+        bc1 = 1 - self.beta1 ** self.step_count
+        bc2 = 1 - self.beta2 ** self.step_count
+        denom = (self.exp_avg_sq / bc2).sqrt_().add_(self.eps)
+        update = (self.exp_avg / bc1) / denom
+        if self.weight_decay:
+            update = update + self.weight_decay * self.flat
+        self.flat.add_(update, alpha=-lr)
+        self.ema.mul_(self.ema_decay).add_(self.flat, alpha=1 - self.ema_decay)
+
+    # ------------------------------------------------------------------
+    def ema_view(self, param: torch.nn.Parameter) -> torch.Tensor:
+        i = self.params.index(param)
+        off = self.offsets[i]
+        return self.ema[off:off + param.numel()].view(param.shape)
+
+    def load_ema_into_params(self):
+        """Swap EMA weights into the live parameters (for eval); returns a
+        tensor holding the previous params so they can be restored."""
+        saved = self.flat.clone()
+        self.flat.copy_(self.ema)
+        return saved
+
+    def restore_params(self, saved: torch.Tensor):
+        self.flat.copy_(saved)
+
+    # ------------------------------------------------------------------
+    def state_dict(self):
+        return {
+            "flat": self.flat, "exp_avg": self.exp_avg,
+            "exp_avg_sq": self.exp_avg_sq, "ema": self.ema,
+            "step_count": self.step_count,
+            "hyper": {"lr": self.lr, "betas": (self.beta1, self.beta2),
+                      "eps": self.eps, "weight_decay": self.weight_decay,
+                      "ema_decay": self.ema_decay},
+        }
+
+    def load_state_dict(self, sd):
+        with torch.no_grad():
+            self.flat.copy_(sd["flat"])
+            self.exp_avg.copy_(sd["exp_avg"])
+            self.exp_avg_sq.copy_(sd["exp_avg_sq"])
+            self.ema.copy_(sd["ema"])
+        self.step_count = int(sd["step_count"])
+
+
+def warmup_cosine_schedule(base_lr: float, warmup_steps: int, total_steps: int,
+                           final_scale: float = 0.0) -> Callable[[int], float]:
+    """Warmup->cosine LR (reference training.py:597-601)."""
+    def sched(step: int) -> float:
+        if warmup_steps > 0 and step < warmup_steps:
+            return base_lr * step / warmup_steps
+        if total_steps <= warmup_steps:
+            return base_lr
+        t = (step - warmup_steps) / max(1, total_steps - warmup_steps)
+        t = min(max(t, 0.0), 1.0)
+        return base_lr * (final_scale + (1 - final_scale) * 0.5 * (1 + math.cos(math.pi * t)))
+    return sched

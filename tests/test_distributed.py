@@ -1,0 +1,100 @@
+"""Multi-process data-parallel tests on gloo (CPU, world_size=2).
+
+Verifies the RCCL-path code (bucketed all-reduce, rank-fold-in RNG, grad
+mean semantics) is correct by construction — the same code runs over
+backend 'nccl' (RCCL) on the GPU nodes.
+"""
+import os
+
+import pytest
+import torch
+import torch.multiprocessing as mp
+
+from flaxdiff_amd.models import Unet
+
+
+def _worker(rank, world, port, results):
+    os.environ.update({
+        "RANK": str(rank), "WORLD_SIZE": str(world), "LOCAL_RANK": str(rank),
+        "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+    })
+    import torch.distributed as dist
+    from flaxdiff_amd.predictors import EpsilonPredictionTransform
+    from flaxdiff_amd.schedulers import CosineNoiseScheduler
+    from flaxdiff_amd.trainer import DiffusionTrainer
+
+    torch.manual_seed(1234)  # same init on both ranks pre-broadcast
+    m = Unet(emb_features=32, feature_depths=[8, 16],
+             attention_configs=[None, {"heads": 2}], num_res_blocks=1,
+             num_middle_res_blocks=1, norm_groups=4, context_dim=16)
+    tr = DiffusionTrainer(m, CosineNoiseScheduler(1000),
+                          EpsilonPredictionTransform(), name=f"ddp",
+                          checkpoint_base_path="/tmp/fdiff_ddp_test",
+                          text_context_shape=(4, 16), distributed=True)
+    # per-rank distinct data
+    g = torch.Generator().manual_seed(100 + rank)
+    batch = {"image": torch.randint(0, 255, (4, 16, 16, 3), generator=g,
+                                    dtype=torch.uint8)}
+    losses = []
+    for _ in range(3):
+        losses.append(tr.train_step(batch)["loss"])
+    # after sync steps params must be identical across ranks
+    flat = tr.optimizer.flat.clone()
+    gathered = [torch.zeros_like(flat) for _ in range(world)]
+    dist.all_gather(gathered, flat)
+    same = all(torch.allclose(gathered[0], gi, atol=1e-6) for gi in gathered)
+    results[rank] = {"losses": losses, "params_equal": bool(same)}
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_data_parallel_two_ranks():
+    world = 2
+    ctx = mp.get_context("spawn")
+    with ctx.Manager() as man:
+        results = man.dict()
+        port = 29731
+        procs = [ctx.Process(target=_worker, args=(r, world, port, results))
+                 for r in range(world)]
+        for p in procs:
+            p.start()
+        for p in procs:
+            p.join(timeout=240)
+        for p in procs:
+            assert p.exitcode == 0, f"worker failed: {p.exitcode}"
+        res = dict(results)
+    assert res[0]["params_equal"] and res[1]["params_equal"]
+    # losses are all-reduced means -> identical across ranks
+    assert res[0]["losses"] == pytest.approx(res[1]["losses"], rel=1e-5)
+
+
+def _rng_worker(rank, world, port, results):
+    os.environ.update({
+        "RANK": str(rank), "WORLD_SIZE": str(world), "LOCAL_RANK": str(rank),
+        "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+    })
+    import torch.distributed as dist
+    from flaxdiff_amd.parallel import init_distributed
+    from flaxdiff_amd.utils import RandomMarkovState
+    init_distributed()
+    st = RandomMarkovState(0).fold_in(rank)
+    _, key = st.get_random_key()
+    results[rank] = float(key.normal((4,)).sum())
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(120)
+def test_per_rank_rng_decorrelated():
+    """fold_in(rank) gives distinct random streams (reference
+    diffusion_trainer.py:158 per-device fold-in)."""
+    ctx = mp.get_context("spawn")
+    with ctx.Manager() as man:
+        results = man.dict()
+        procs = [ctx.Process(target=_rng_worker, args=(r, 2, 29732, results))
+                 for r in range(2)]
+        for p in procs:
+            p.start()
+        for p in procs:
+            p.join(timeout=100)
+        res = dict(results)
+    assert res[0] != res[1]
