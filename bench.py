@@ -1,0 +1,131 @@
+#!/usr/bin/env python3
+"""Flagship training benchmark — the driver contract.
+
+Measures the BASELINE.json metric: train images/sec/node for the 64x64
+unconditional EDM UNet (feature_depths [64,128,256,512], heads 4, 2 res
+blocks — README.md:363-367 of the reference) at global batch 256, bf16,
+synthetic data, random-init weights. N>1 runs one rank per GPU over RCCL
+(launched by torch.distributed.run); global batch stays 256 (strong scaling,
+BASELINE config 3).
+
+Usage: python bench.py [--gpus N] [--steps K] [--warmup W]
+"""
+import argparse
+import json
+import os
+import time
+
+import torch
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--global-batch", type=int, default=256)
+    ap.add_argument("--resolution", type=int, default=64)
+    ap.add_argument("--dtype", type=str, default="bf16", choices=["bf16", "fp32"])
+    args = ap.parse_args()
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    n_gpus = max(world, 1)
+    if world <= 1:
+        n_gpus = 1
+
+    use_gpu = torch.cuda.is_available()
+    compute_dtype = torch.bfloat16 if (args.dtype == "bf16" and use_gpu) else torch.float32
+
+    from flaxdiff_amd import parallel
+    from flaxdiff_amd.models import Unet
+    from flaxdiff_amd.predictors import KarrasPredictionTransform
+    from flaxdiff_amd.schedulers import EDMNoiseScheduler
+    from flaxdiff_amd.trainer import DiffusionTrainer
+
+    torch.manual_seed(1234)
+    model = Unet(
+        output_channels=3,
+        emb_features=256,
+        feature_depths=[64, 128, 256, 512],
+        attention_configs=[{"heads": 4}] * 4,
+        num_res_blocks=2,
+        num_middle_res_blocks=1,
+        norm_groups=8,
+        context_dim=768,
+    )
+    trainer = DiffusionTrainer(
+        model,
+        EDMNoiseScheduler(1, sigma_max=80, sigma_data=0.5),
+        KarrasPredictionTransform(sigma_data=0.5),
+        name="bench",
+        checkpoint_base_path="/tmp/fdiff_bench_ckpt",
+        compute_dtype=compute_dtype,
+        distributed=(world > 1),
+        optimizer_kwargs={"lr": 2.7e-4},
+    )
+    dev = trainer.device
+
+    local_batch = max(args.global_batch // n_gpus, 1)
+    g = torch.Generator().manual_seed(42 + rank)
+    batch = {
+        "image": torch.randint(0, 255, (local_batch, args.resolution,
+                                        args.resolution, 3),
+                               generator=g, dtype=torch.uint8).to(dev),
+    }
+
+    # warmup
+    for _ in range(args.warmup):
+        trainer.train_step(batch)
+
+    parallel.barrier()
+    if use_gpu:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    loss = 0.0
+    for _ in range(args.steps):
+        loss = trainer.train_step(batch)["loss"]
+    parallel.barrier()
+    if use_gpu:
+        torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks
+    el = torch.tensor([elapsed], dtype=torch.float64)
+    if world > 1:
+        import torch.distributed as dist
+        el_dev = el.to(dev) if use_gpu else el
+        dist.all_reduce(el_dev, op=dist.ReduceOp.MAX)
+        elapsed = float(el_dev.item())
+
+    images_total = args.global_batch if world > 1 else local_batch
+    value = images_total * args.steps / elapsed
+    ms_per_step = elapsed / args.steps * 1000
+
+    if rank == 0:
+        print(json.dumps({
+            "metric": "train images/sec/node (64x64 uncond EDM UNet, bs=256)",
+            "value": value,
+            "unit": "images/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "strong",
+            "vs_baseline": None,
+            "dtype": "bf16" if compute_dtype == torch.bfloat16 else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": "unet_64px_[64,128,256,512]_res2_heads4",
+                "global_batch": images_total,
+                "resolution": args.resolution,
+                "parallelism": f"dp{n_gpus}",
+                "schedule": "EDM + KarrasPredictionTransform(sigma_data=0.5)",
+                "final_loss": loss,
+            },
+        }))
+
+
+if __name__ == "__main__":
+    main()

@@ -125,6 +125,43 @@ def rms_norm(x: torch.Tensor, gamma: torch.Tensor, eps: float = 1e-5) -> torch.T
 # Conv2d NHWC (3x3 SAME / 1x1 / stride-2) — LDS-tiled implicit GEMM on MFMA
 # ---------------------------------------------------------------------------
 
+def _conv2d_wgrad_gemms(dy: torch.Tensor, x: torch.Tensor, kh: int, kw: int,
+                        stride: int) -> torch.Tensor:
+    """Weight gradient as KH*KW shifted-view GEMMs on MFMA (hipBLASLt).
+
+    dw[r,s,ci,co] = sum_m x[b, oh*st+r-pt, ow*st+s-pl, ci] * dy[b,oh,ow,co]
+    over the in-bounds output positions. Each (r,s) tap is one [Ci x M']@[M' x Co]
+    library GEMM over a strided view — no im2col materialization. A fully
+    hand-written LDS-tiled wgrad kernel is the planned replacement.
+    """
+    B, H, W, Ci = x.shape
+    _, OH, OW, Co = dy.shape
+    pt = max((OH - 1) * stride + kh - H, 0) // 2
+    pl = max((OW - 1) * stride + kw - W, 0) // 2
+    dw = torch.zeros(kh, kw, Ci, Co, dtype=torch.float32, device=x.device)
+    for r in range(kh):
+        oh_lo = max(0, -((pt - r) // -stride))  # ceil((pt-r)/stride), clamped
+        oh_hi = min(OH - 1, (H - 1 - r + pt) // stride)
+        if oh_hi < oh_lo:
+            continue
+        ih_lo = oh_lo * stride + r - pt
+        noh = oh_hi - oh_lo + 1
+        for s in range(kw):
+            ow_lo = max(0, -((pl - s) // -stride))
+            ow_hi = min(OW - 1, (W - 1 - s + pl) // stride)
+            if ow_hi < ow_lo:
+                continue
+            iw_lo = ow_lo * stride + s - pl
+            now = ow_hi - ow_lo + 1
+            x_sub = x[:, ih_lo:ih_lo + (noh - 1) * stride + 1:stride,
+                      iw_lo:iw_lo + (now - 1) * stride + 1:stride, :]
+            dy_sub = dy[:, oh_lo:oh_hi + 1, ow_lo:ow_hi + 1, :]
+            a = x_sub.reshape(-1, Ci)
+            b = dy_sub.reshape(-1, Co)
+            dw[r, s] += (a.transpose(0, 1) @ b).float()
+    return dw
+
+
 class _Conv2dFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, w, b, stride):
@@ -144,7 +181,7 @@ class _Conv2dFn(torch.autograd.Function):
             if ctx.needs_input_grad[0] else None
         dw = None
         if ctx.needs_input_grad[1]:
-            dw = ext.conv2d_wgrad(dy, x, w.shape[0], w.shape[1], ctx.stride)
+            dw = _conv2d_wgrad_gemms(dy, x, w.shape[0], w.shape[1], ctx.stride)
             dw = dw.to(w.dtype)
         db = dy.reshape(-1, dy.shape[-1]).sum(0).to(w.dtype) if ctx.has_bias else None
         return dx, dw, db, None

@@ -1,0 +1,330 @@
+"""GPU numerics tests: every HIP kernel vs the fp32 torch reference
+(CPU reference path is the oracle, bit-tolerance asserts).
+All tests are @pytest.mark.gpu and run on the MI355X box.
+"""
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+if torch.cuda.is_available():
+    from flaxdiff_amd import ops
+    from flaxdiff_amd.ops import reference
+
+
+def _dev():
+    return torch.device("cuda:0")
+
+
+def rel_err(a, b):
+    return ((a.float() - b.float()).abs().max() /
+            (b.float().abs().max() + 1e-6)).item()
+
+
+@pytest.fixture(autouse=True)
+def _require_hip():
+    assert ops.hip_available(), "HIP extension must be loaded on GPU"
+
+
+# ---------------------------------------------------------------------------
+# elementwise
+# ---------------------------------------------------------------------------
+
+def test_fwd_diffusion_bf16():
+    g = torch.Generator(device="cuda").manual_seed(0)
+    x0 = torch.randn(4, 8, 8, 3, device=_dev(), generator=g, dtype=torch.float32)
+    eps = torch.randn(4, 8, 8, 3, device=_dev(), generator=g, dtype=torch.float32)
+    a = torch.rand(4, 1, 1, 1, device=_dev(), generator=g)
+    s = torch.rand(4, 1, 1, 1, device=_dev(), generator=g)
+    ref = reference.forward_diffusion(x0, eps, a, s)
+    out = ops.forward_diffusion(x0.bfloat16(), eps.bfloat16(), a, s)
+    assert rel_err(out, ref) < 2e-2
+
+
+def test_upsample2x_roundtrip():
+    x = torch.randn(2, 8, 8, 16, device=_dev()).bfloat16()
+    y = ops.nearest_upsample_2x(x)
+    ref = reference.nearest_upsample_2x_nhwc(x.float().cpu())
+    assert torch.allclose(y.float().cpu(), ref, atol=1e-2)
+    # backward: sum over 2x2 windows
+    xx = x.float().requires_grad_(True)
+    yy = reference.nearest_upsample_2x_nhwc(xx)
+    dy = torch.randn_like(yy)
+    yy.backward(dy)
+    from flaxdiff_amd.ops import _require_ext
+    dx = _require_ext().upsample2x_bwd(dy.bfloat16().contiguous())
+    assert rel_err(dx, xx.grad) < 2e-2
+
+
+# ---------------------------------------------------------------------------
+# GroupNorm + SiLU
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize("silu", [False, True])
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_gn_silu_fwd(silu, dtype):
+    torch.manual_seed(0)
+    B, H, W, C, G = 4, 16, 16, 96, 8
+    x = torch.randn(B, H, W, C, device=_dev(), dtype=torch.float32)
+    gamma = torch.randn(C, device=_dev())
+    beta = torch.randn(C, device=_dev())
+    ref = reference.group_norm_nhwc(x.cpu(), G, gamma.cpu(), beta.cpu(), 1e-5, silu)
+    y = ops.group_norm(x.to(dtype), G, gamma.to(dtype), beta.to(dtype), 1e-5, silu)
+    tol = 1e-4 if dtype == torch.float32 else 3e-2
+    assert rel_err(y.cpu(), ref) < tol
+
+
+@pytest.mark.parametrize("silu", [False, True])
+def test_gn_silu_bwd(silu):
+    torch.manual_seed(1)
+    B, H, W, C, G = 2, 8, 8, 32, 4
+    x_cpu = torch.randn(B, H, W, C, dtype=torch.float64)
+    gamma_cpu = torch.randn(C, dtype=torch.float64)
+    beta_cpu = torch.randn(C, dtype=torch.float64)
+    x_ref = x_cpu.clone().requires_grad_(True)
+    g_ref = gamma_cpu.clone().requires_grad_(True)
+    b_ref = beta_cpu.clone().requires_grad_(True)
+    y_ref = reference.group_norm_nhwc(x_ref, G, g_ref, b_ref, 1e-5, silu)
+    dy = torch.randn_like(y_ref)
+    y_ref.backward(dy)
+
+    x = x_cpu.float().to(_dev()).requires_grad_(True)
+    g = gamma_cpu.float().to(_dev()).requires_grad_(True)
+    b = beta_cpu.float().to(_dev()).requires_grad_(True)
+    y = ops.group_norm(x, G, g, b, 1e-5, silu)
+    y.backward(dy.float().to(_dev()))
+
+    assert rel_err(x.grad.cpu(), x_ref.grad) < 1e-3
+    assert rel_err(g.grad.cpu(), g_ref.grad) < 1e-3
+    assert rel_err(b.grad.cpu(), b_ref.grad) < 1e-3
+
+
+# ---------------------------------------------------------------------------
+# RMSNorm
+# ---------------------------------------------------------------------------
+
+def test_rms_norm_fwd_bwd():
+    torch.manual_seed(2)
+    R, C = 64, 128
+    x_cpu = torch.randn(R, C, dtype=torch.float64)
+    g_cpu = torch.randn(C, dtype=torch.float64)
+    x_ref = x_cpu.clone().requires_grad_(True)
+    g_ref = g_cpu.clone().requires_grad_(True)
+    y_ref = reference.rms_norm(x_ref, g_ref, 1e-5)
+    dy = torch.randn_like(y_ref)
+    y_ref.backward(dy)
+
+    x = x_cpu.float().to(_dev()).requires_grad_(True)
+    g = g_cpu.float().to(_dev()).requires_grad_(True)
+    y = ops.rms_norm(x, g, 1e-5)
+    assert rel_err(y.cpu(), y_ref) < 1e-3
+    y.backward(dy.float().to(_dev()))
+    assert rel_err(x.grad.cpu(), x_ref.grad) < 1e-3
+    assert rel_err(g.grad.cpu(), g_ref.grad) < 1e-3
+
+
+# ---------------------------------------------------------------------------
+# conv2d implicit GEMM
+# ---------------------------------------------------------------------------
+
+CONV_SHAPES = [
+    # B, H, W, Ci, Co, k, stride — the UNet channel zoo (incl. odd concats)
+    (2, 16, 16, 64, 64, 3, 1),
+    (2, 16, 16, 3, 64, 3, 1),        # stem: Ci=3
+    (2, 16, 16, 64, 3, 3, 1),        # head: Co=3
+    (2, 16, 16, 128, 256, 3, 1),
+    (2, 16, 16, 192, 256, 3, 1),     # concat channels
+    (2, 16, 16, 64, 128, 3, 2),      # downsample
+    (2, 15, 15, 32, 48, 3, 2),       # odd spatial
+    (1, 8, 8, 768, 512, 3, 1),       # decoder concat
+]
+
+
+@pytest.mark.parametrize("B,H,W,Ci,Co,k,st", CONV_SHAPES)
+def test_conv2d_fwd(B, H, W, Ci, Co, k, st):
+    torch.manual_seed(3)
+    x = torch.randn(B, H, W, Ci) * 0.5
+    w = torch.randn(k, k, Ci, Co) * (1.0 / (k * k * Ci) ** 0.5)
+    bias = torch.randn(Co) * 0.1
+    ref = reference.conv2d_nhwc(x, w, bias, stride=st, padding="same")
+    y = ops.conv2d(x.bfloat16().to(_dev()), w.bfloat16().to(_dev()),
+                   bias.bfloat16().to(_dev()), stride=st)
+    assert y.shape == ref.shape
+    err = rel_err(y.cpu(), ref)
+    assert err < 3e-2, f"conv fwd rel err {err}"
+
+
+@pytest.mark.parametrize("B,H,W,Ci,Co,k,st", CONV_SHAPES[:6])
+def test_conv2d_backward(B, H, W, Ci, Co, k, st):
+    torch.manual_seed(4)
+    x_cpu = torch.randn(B, H, W, Ci) * 0.5
+    w_cpu = torch.randn(k, k, Ci, Co) * (1.0 / (k * k * Ci) ** 0.5)
+    b_cpu = torch.randn(Co) * 0.1
+
+    x_ref = x_cpu.clone().requires_grad_(True)
+    w_ref = w_cpu.clone().requires_grad_(True)
+    b_ref = b_cpu.clone().requires_grad_(True)
+    y_ref = reference.conv2d_nhwc(x_ref, w_ref, b_ref, stride=st, padding="same")
+    dy = torch.randn_like(y_ref)
+    y_ref.backward(dy)
+
+    x = x_cpu.bfloat16().to(_dev()).requires_grad_(True)
+    w = w_cpu.bfloat16().to(_dev()).requires_grad_(True)
+    b = b_cpu.bfloat16().to(_dev()).requires_grad_(True)
+    y = ops.conv2d(x, w, b, stride=st)
+    y.backward(dy.bfloat16().to(_dev()))
+
+    assert rel_err(x.grad.cpu(), x_ref.grad) < 5e-2
+    assert rel_err(w.grad.cpu(), w_ref.grad) < 5e-2
+    assert rel_err(b.grad.cpu(), b_ref.grad) < 5e-2
+
+
+# ---------------------------------------------------------------------------
+# attention
+# ---------------------------------------------------------------------------
+
+ATTN_SHAPES = [
+    # B, H, Sq, Skv, D
+    (2, 4, 256, 77, 16),     # UNet cross-attn level 0 shape class
+    (2, 4, 64, 77, 64),
+    (1, 4, 128, 77, 128),
+    (1, 2, 256, 256, 64),    # self-attention
+    (1, 2, 300, 300, 32),    # non-multiple seq lens
+    (1, 1, 64, 1024, 64),    # long KV (tiled online softmax)
+]
+
+
+@pytest.mark.parametrize("B,H,Sq,Skv,D", ATTN_SHAPES)
+def test_attn_fwd(B, H, Sq, Skv, D):
+    torch.manual_seed(5)
+    q = torch.randn(B, H, Sq, D)
+    k = torch.randn(B, H, Skv, D)
+    v = torch.randn(B, H, Skv, D)
+    ref = reference.attention(q, k, v)
+    o = ops.attention(q.bfloat16().to(_dev()), k.bfloat16().to(_dev()),
+                      v.bfloat16().to(_dev()))
+    err = rel_err(o.cpu(), ref)
+    assert err < 3e-2, f"attn fwd rel err {err}"
+
+
+def test_attn_spiked_softmax():
+    """Outlier K row forces large max shifts (rule 26-style branch test)."""
+    torch.manual_seed(6)
+    q = torch.randn(1, 1, 64, 64)
+    k = torch.randn(1, 1, 128, 64)
+    k[0, 0, 100] = q[0, 0, 5] * 10  # spike late tile
+    v = torch.randn(1, 1, 128, 64)
+    ref = reference.attention(q, k, v)
+    o = ops.attention(q.bfloat16().to(_dev()), k.bfloat16().to(_dev()),
+                      v.bfloat16().to(_dev()))
+    assert rel_err(o.cpu(), ref) < 3e-2
+
+
+def test_attn_backward_composed():
+    torch.manual_seed(7)
+    B, H, Sq, Skv, D = 1, 2, 64, 77, 32
+    q_cpu = torch.randn(B, H, Sq, D)
+    k_cpu = torch.randn(B, H, Skv, D)
+    v_cpu = torch.randn(B, H, Skv, D)
+    qr = q_cpu.clone().requires_grad_(True)
+    kr = k_cpu.clone().requires_grad_(True)
+    vr = v_cpu.clone().requires_grad_(True)
+    o_ref = reference.attention(qr, kr, vr)
+    do = torch.randn_like(o_ref)
+    o_ref.backward(do)
+
+    q = q_cpu.bfloat16().to(_dev()).requires_grad_(True)
+    k = k_cpu.bfloat16().to(_dev()).requires_grad_(True)
+    v = v_cpu.bfloat16().to(_dev()).requires_grad_(True)
+    o = ops.attention(q, k, v)
+    o.backward(do.bfloat16().to(_dev()))
+    assert rel_err(q.grad.cpu(), qr.grad) < 6e-2
+    assert rel_err(k.grad.cpu(), kr.grad) < 6e-2
+    assert rel_err(v.grad.cpu(), vr.grad) < 6e-2
+
+
+# ---------------------------------------------------------------------------
+# fused AdamW + EMA
+# ---------------------------------------------------------------------------
+
+def test_fused_adamw_ema_matches_cpu():
+    torch.manual_seed(8)
+    n = 10_000
+    p = torch.randn(n)
+    g = torch.randn(n)
+    m = torch.zeros(n)
+    v = torch.zeros(n)
+    ema = p.clone()
+
+    # CPU reference (same math as FlatAdamWEMA CPU path)
+    lr, b1, b2, eps, wd, ema_d, gs = 1e-3, 0.9, 0.999, 1e-8, 0.01, 0.999, 0.5
+    for step in (1, 2, 3):
+        gg = g * gs
+        m_ref = m.clone()
+        v_ref = v.clone()
+    # do it properly sequentially
+    p_ref, m_ref, v_ref, ema_ref = p.clone(), m.clone(), v.clone(), ema.clone()
+    for step in (1, 2, 3):
+        gg = g * gs
+        m_ref = b1 * m_ref + (1 - b1) * gg
+        v_ref = b2 * v_ref + (1 - b2) * gg * gg
+        mhat = m_ref / (1 - b1 ** step)
+        vhat = v_ref / (1 - b2 ** step)
+        upd = mhat / (vhat.sqrt() + eps) + wd * p_ref
+        p_ref = p_ref - lr * upd
+        ema_ref = ema_d * ema_ref + (1 - ema_d) * p_ref
+
+    dev = _dev()
+    pg, gg_, mg, vg, eg = (t.to(dev).contiguous() for t in (p, g, m, v, ema))
+    for step in (1, 2, 3):
+        ops.fused_adamw_ema(pg, gg_, mg, vg, eg, None, lr=lr, beta1=b1, beta2=b2,
+                            eps=eps, weight_decay=wd, step=step, ema_decay=ema_d,
+                            grad_scale=gs)
+    assert rel_err(pg.cpu(), p_ref) < 1e-4
+    assert rel_err(eg.cpu(), ema_ref) < 1e-4
+    assert rel_err(mg.cpu(), m_ref) < 1e-4
+    assert rel_err(vg.cpu(), v_ref) < 1e-4
+
+
+# ---------------------------------------------------------------------------
+# end-to-end model step on GPU
+# ---------------------------------------------------------------------------
+
+def test_unet_train_step_bf16():
+    from flaxdiff_amd.models import Unet
+    from flaxdiff_amd.predictors import KarrasPredictionTransform
+    from flaxdiff_amd.schedulers import EDMNoiseScheduler
+    from flaxdiff_amd.trainer import DiffusionTrainer
+    torch.manual_seed(0)
+    m = Unet(emb_features=128, feature_depths=[32, 64],
+             attention_configs=[{"heads": 4}, {"heads": 4}], num_res_blocks=2,
+             num_middle_res_blocks=1, norm_groups=8, context_dim=768)
+    tr = DiffusionTrainer(m, EDMNoiseScheduler(1, sigma_max=80),
+                          KarrasPredictionTransform(sigma_data=0.5),
+                          name="gputest", checkpoint_base_path="/tmp/fdiff_gputest",
+                          compute_dtype=torch.bfloat16, distributed=False)
+    batch = {"image": torch.randint(0, 255, (4, 32, 32, 3), dtype=torch.uint8)}
+    losses = [tr.train_step(batch)["loss"] for _ in range(5)]
+    assert all(l == l for l in losses), f"NaN loss: {losses}"
+
+
+def test_sampling_on_gpu():
+    from flaxdiff_amd.models import Unet
+    from flaxdiff_amd.predictors import KarrasPredictionTransform
+    from flaxdiff_amd.samplers import HeunSampler
+    from flaxdiff_amd.schedulers import KarrasVENoiseScheduler
+    torch.manual_seed(0)
+    m = Unet(emb_features=128, feature_depths=[32, 64],
+             attention_configs=[{"heads": 4}, {"heads": 4}], num_res_blocks=1,
+             num_middle_res_blocks=1, norm_groups=8, context_dim=768).to(_dev())
+    ns = KarrasVENoiseScheduler(timesteps=1000, sigma_data=0.5)
+    sampler = HeunSampler(
+        model=lambda x, t, *c: m(x.bfloat16(), t, *(ci.bfloat16() for ci in c)),
+        noise_schedule=ns,
+        model_output_transform=KarrasPredictionTransform(sigma_data=0.5))
+    ctx = torch.zeros(2, 77, 768, device=_dev())
+    out = sampler.generate_samples(num_samples=2, resolution=32, diffusion_steps=5,
+                                   model_conditioning_inputs=(ctx,),
+                                   device=_dev(), dtype=torch.bfloat16)
+    assert out.shape == (2, 32, 32, 3)
+    assert torch.isfinite(out.float()).all()
