@@ -23,12 +23,22 @@ import torch
 import torch.distributed as dist
 
 
+def _default_device() -> torch.device:
+    return torch.device("cuda:0") if torch.cuda.is_available() else torch.device("cpu")
+
+
 @dataclass
 class DistContext:
     rank: int = 0
     world_size: int = 1
     local_rank: int = 0
-    device: torch.device = torch.device("cpu")
+    device: torch.device = None  # type: ignore[assignment]
+
+    def __post_init__(self):
+        if self.device is None:
+            self.device = _default_device()
+            if self.device.type == "cuda":
+                torch.cuda.set_device(self.device)
 
     @property
     def is_distributed(self) -> bool:

@@ -303,6 +303,8 @@ def test_unet_train_step_bf16():
                           KarrasPredictionTransform(sigma_data=0.5),
                           name="gputest", checkpoint_base_path="/tmp/fdiff_gputest",
                           compute_dtype=torch.bfloat16, distributed=False)
+    assert tr.device.type == "cuda", "trainer must run on the GPU"
+    assert next(tr.model.parameters()).is_cuda
     batch = {"image": torch.randint(0, 255, (4, 32, 32, 3), dtype=torch.uint8)}
     losses = [tr.train_step(batch)["loss"] for _ in range(5)]
     assert all(l == l for l in losses), f"NaN loss: {losses}"
