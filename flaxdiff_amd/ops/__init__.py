@@ -181,7 +181,7 @@ class _Conv2dFn(torch.autograd.Function):
             if ctx.needs_input_grad[0] else None
         dw = None
         if ctx.needs_input_grad[1]:
-            dw = _conv2d_wgrad_gemms(dy, x, w.shape[0], w.shape[1], ctx.stride)
+            dw = ext.conv2d_wgrad(dy, x, w.shape[0], w.shape[1], ctx.stride)
             dw = dw.to(w.dtype)
         db = dy.reshape(-1, dy.shape[-1]).sum(0).to(w.dtype) if ctx.has_bias else None
         return dx, dw, db, None
