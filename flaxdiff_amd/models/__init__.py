@@ -5,6 +5,11 @@ from .common import (Conv, ConvLayer, Dense, Downsample, FourierEmbedding,
 from .attention import (BasicTransformerBlock, EfficientAttention, FeedForward,
                         GEGLU, NormalAttention, TransformerBlock)
 from .unet import Unet
+from .simple_dit import DiTBlock, SimpleDiT
+from .simple_vit import SimpleUDiT, UViT
+from .vit_common import (AdaLNParams, AdaLNZero, PatchEmbedding,
+                         PositionalEncoding, RoPEAttention, RotaryEmbedding,
+                         apply_rotary_embedding)
 
 __all__ = [
     "Unet", "Conv", "ConvLayer", "Dense", "Downsample", "FourierEmbedding",
@@ -12,4 +17,7 @@ __all__ = [
     "TimeEmbedding", "TimeProjection", "Upsample", "WeightStandardizedConv",
     "BasicTransformerBlock", "EfficientAttention", "FeedForward", "GEGLU",
     "NormalAttention", "TransformerBlock",
+    "DiTBlock", "SimpleDiT", "SimpleUDiT", "UViT",
+    "AdaLNParams", "AdaLNZero", "PatchEmbedding", "PositionalEncoding",
+    "RoPEAttention", "RotaryEmbedding", "apply_rotary_embedding",
 ]
