@@ -6,6 +6,8 @@ from .attention import (BasicTransformerBlock, EfficientAttention, FeedForward,
                         GEGLU, NormalAttention, TransformerBlock)
 from .unet import Unet
 from .simple_dit import DiTBlock, SimpleDiT
+from .simple_mmdit import (HierarchicalMMDiT, MMAdaLNZero, MMDiTBlock,
+                           PatchExpanding, PatchMerging, SimpleMMDiT)
 from .simple_vit import SimpleUDiT, UViT
 from .vit_common import (AdaLNParams, AdaLNZero, PatchEmbedding,
                          PositionalEncoding, RoPEAttention, RotaryEmbedding,
@@ -18,6 +20,8 @@ __all__ = [
     "BasicTransformerBlock", "EfficientAttention", "FeedForward", "GEGLU",
     "NormalAttention", "TransformerBlock",
     "DiTBlock", "SimpleDiT", "SimpleUDiT", "UViT",
+    "HierarchicalMMDiT", "MMAdaLNZero", "MMDiTBlock", "PatchExpanding",
+    "PatchMerging", "SimpleMMDiT",
     "AdaLNParams", "AdaLNZero", "PatchEmbedding", "PositionalEncoding",
     "RoPEAttention", "RotaryEmbedding", "apply_rotary_embedding",
 ]

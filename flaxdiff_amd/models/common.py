@@ -44,10 +44,32 @@ def variance_scaling_(tensor: torch.Tensor, fan_in: int, fan_out: int,
     return tensor
 
 
+class _ShadowCast(torch.autograd.Function):
+    """fp32 master param -> its pre-computed bf16 shadow (zero-copy forward).
+
+    The shadow view is refreshed by the fused Adam kernel each optimizer step
+    (trainer/optim.py), so the forward is a no-op instead of a cast kernel;
+    the backward casts the incoming bf16 grad to fp32 so autograd accumulates
+    into the flat fp32 grad buffer.
+    """
+
+    @staticmethod
+    def forward(ctx, p):
+        return p._shadow_bf16
+
+    @staticmethod
+    def backward(ctx, dy):
+        return dy.float()
+
+
 def _cast(p: Optional[torch.Tensor], dtype) -> Optional[torch.Tensor]:
     if p is None:
         return None
-    return p if p.dtype == dtype else p.to(dtype)
+    if p.dtype == dtype:
+        return p
+    if dtype == torch.bfloat16 and getattr(p, "_shadow_bf16", None) is not None:
+        return _ShadowCast.apply(p)
+    return p.to(dtype)
 
 
 # ---------------------------------------------------------------------------
