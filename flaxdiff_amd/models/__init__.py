@@ -9,6 +9,8 @@ from .simple_dit import DiTBlock, SimpleDiT
 from .simple_mmdit import (HierarchicalMMDiT, MMAdaLNZero, MMDiTBlock,
                            PatchExpanding, PatchMerging, SimpleMMDiT)
 from .simple_vit import SimpleUDiT, UViT
+from .ssm_dit import (BidirectionalS5Layer, HybridSSMAttentionDiT, S5Layer,
+                      SSMDiTBlock, SpatialFusionConv)
 from .vit_common import (AdaLNParams, AdaLNZero, PatchEmbedding,
                          PositionalEncoding, RoPEAttention, RotaryEmbedding,
                          apply_rotary_embedding)
@@ -22,6 +24,8 @@ __all__ = [
     "DiTBlock", "SimpleDiT", "SimpleUDiT", "UViT",
     "HierarchicalMMDiT", "MMAdaLNZero", "MMDiTBlock", "PatchExpanding",
     "PatchMerging", "SimpleMMDiT",
+    "BidirectionalS5Layer", "HybridSSMAttentionDiT", "S5Layer", "SSMDiTBlock",
+    "SpatialFusionConv",
     "AdaLNParams", "AdaLNZero", "PatchEmbedding", "PositionalEncoding",
     "RoPEAttention", "RotaryEmbedding", "apply_rotary_embedding",
 ]
