@@ -1,0 +1,16 @@
+from .sources import (AUGMENT_MODE_ENV, DataAugmenter, DataSource,
+                      HFDatasetSource, ImageAugmenter, ImageFolderSource,
+                      SyntheticImageSource, TensorSource, datasetMap,
+                      register_dataset, register_image_folder)
+from .dataloaders import (AugmentedDataset, PrefetchLoader, ShardedSampler,
+                          collate_image_batch, get_dataset, get_dataset_online,
+                          make_dataloader)
+
+__all__ = [
+    "AUGMENT_MODE_ENV", "DataAugmenter", "DataSource", "HFDatasetSource",
+    "ImageAugmenter", "ImageFolderSource", "SyntheticImageSource",
+    "TensorSource", "datasetMap", "register_dataset", "register_image_folder",
+    "AugmentedDataset", "PrefetchLoader", "ShardedSampler",
+    "collate_image_batch", "get_dataset", "get_dataset_online",
+    "make_dataloader",
+]
