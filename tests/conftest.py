@@ -4,6 +4,7 @@ import torch
 
 def pytest_configure(config):
     config.addinivalue_line("markers", "gpu: requires an MI355X GPU (run with -m gpu)")
+    config.addinivalue_line("markers", "slow: long-running CPU test")
 
 
 def pytest_collection_modifyitems(config, items):
