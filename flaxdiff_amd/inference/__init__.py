@@ -1,0 +1,6 @@
+from .pipeline import DiffusionInferencePipeline
+from .utils import (canonicalize_architecture, load_from_checkpoint,
+                    map_nested_config, parse_config)
+
+__all__ = ["DiffusionInferencePipeline", "canonicalize_architecture",
+           "load_from_checkpoint", "map_nested_config", "parse_config"]

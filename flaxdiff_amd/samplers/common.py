@@ -189,7 +189,15 @@ class DiffusionSampler:
             model_conditioning_inputs = self.input_config.encode_conditions(
                 conditioning, device=device, dtype=dtype)
         if model_conditioning_inputs is None:
-            model_conditioning_inputs = ()
+            if self.unconditionals:
+                # unconditional generation of a conditional model: feed the
+                # null embeddings (reference samplers/common.py:315-349)
+                model_conditioning_inputs = tuple(
+                    torch.as_tensor(u, device=device).unsqueeze(0)
+                    .expand(num_samples, *u.shape).to(dtype)
+                    for u in self.unconditionals)
+            else:
+                model_conditioning_inputs = ()
 
         def sample_model_fn(x_t, t, *cond):
             return self.sample_model(x_t, t, *cond)
