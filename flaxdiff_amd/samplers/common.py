@@ -45,6 +45,19 @@ class DiffusionSampler:
             self.min_inv_rho = noise_schedule.min_inv_rho
             self.max_inv_rho = noise_schedule.max_inv_rho
 
+        self._graph_capture = False
+        self._graphed: Optional["_GraphedSampleModel"] = None
+
+    def enable_graph_capture(self, enabled: bool = True):
+        """Capture the per-step CFG-doubled model forward in a hipGraph and
+        replay it each step (timesteps fed through a device buffer). The
+        per-sampler take_next_step AXPYs stay eager — they are a few
+        elementwise kernels on the sample tensor. No-op without a GPU."""
+        self._graph_capture = enabled and torch.cuda.is_available()
+        if not enabled:
+            self._graphed = None
+        return self
+
     # ------------------------------------------------------------------
     # model evaluation (CFG batch doubling: reference common.py:60-103)
     # ------------------------------------------------------------------
@@ -199,8 +212,17 @@ class DiffusionSampler:
             else:
                 model_conditioning_inputs = ()
 
-        def sample_model_fn(x_t, t, *cond):
-            return self.sample_model(x_t, t, *cond)
+        if self._graph_capture and samples.is_cuda:
+            if (self._graphed is None
+                    or not self._graphed.matches(samples, model_conditioning_inputs)):
+                self._graphed = _GraphedSampleModel(self, samples,
+                                                   model_conditioning_inputs)
+
+            def sample_model_fn(x_t, t, *cond):
+                return self._graphed(x_t, t)
+        else:
+            def sample_model_fn(x_t, t, *cond):
+                return self.sample_model(x_t, t, *cond)
 
         steps = steps_override if steps_override is not None else \
             self.get_steps(start_step, end_step, diffusion_steps)
@@ -247,3 +269,48 @@ class DiffusionSampler:
         else:
             shape = (batch_size, image_size, image_size, image_channels)
         return key.normal(shape, device=device).to(dtype) * variance
+
+
+class _GraphedSampleModel:
+    """hipGraph-captured sample_model: static input/cond/output buffers, one
+    graph replay per model evaluation (SURVEY.md §2.5 MI355X plan).
+
+    Conditioning tensors are copied into static buffers ONCE at capture (they
+    are constant across the sampling loop); x_t and t stream through static
+    buffers each call.
+    """
+
+    def __init__(self, sampler: DiffusionSampler, x: torch.Tensor,
+                 conds: Tuple[torch.Tensor, ...]):
+        self.sampler = sampler
+        self.shape = tuple(x.shape)
+        self.dtype = x.dtype
+        dev = x.device
+        self.static_x = torch.zeros_like(x)
+        self.static_t = torch.zeros(x.shape[0], device=dev, dtype=torch.float32)
+        self.static_conds = tuple(c.clone() for c in conds)
+
+        torch.cuda.synchronize()
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            for _ in range(2):  # warmup allocations outside the graph
+                sampler.sample_model(self.static_x, self.static_t,
+                                     *self.static_conds)
+        torch.cuda.current_stream().wait_stream(side)
+        torch.cuda.synchronize()
+
+        self.graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph):
+            self.out_x0, self.out_eps, self.out_raw = sampler.sample_model(
+                self.static_x, self.static_t, *self.static_conds)
+
+    def matches(self, x: torch.Tensor, conds) -> bool:
+        return tuple(x.shape) == self.shape and x.dtype == self.dtype and \
+            len(conds) == len(self.static_conds)
+
+    def __call__(self, x_t: torch.Tensor, t: torch.Tensor):
+        self.static_x.copy_(x_t)
+        self.static_t.copy_(t)
+        self.graph.replay()
+        return self.out_x0.clone(), self.out_eps.clone(), self.out_raw
