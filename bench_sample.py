@@ -42,8 +42,12 @@ def main():
     schedule = KarrasVENoiseScheduler(1, sigma_max=80, rho=7, sigma_data=0.5)
     transform = KarrasPredictionTransform(sigma_data=0.5)
 
+    null_ctx = torch.zeros(1, 77, 768, device=dev)
+
     def bench(sampler_cls, graph: bool, label: str):
-        sampler = sampler_cls(model=lambda x, t, *c: model(x.to(dtype), t).float(),
+        sampler = sampler_cls(model=lambda x, t, *c: model(
+                                  x.to(dtype), t,
+                                  null_ctx.expand(x.shape[0], -1, -1).to(dtype)).float(),
                               noise_schedule=schedule,
                               model_output_transform=transform,
                               guidance_scale=args.guidance,

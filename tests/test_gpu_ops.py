@@ -349,9 +349,13 @@ def test_graph_captured_sampling_matches_eager():
     schedule = KarrasVENoiseScheduler(1, sigma_max=80, rho=7, sigma_data=0.5)
     transform = KarrasPredictionTransform(sigma_data=0.5)
 
+    null_ctx = torch.zeros(1, 77, 768, device="cuda")
+
     def run(graph):
         s = EulerAncestralSampler(
-            model=lambda x, t, *c: model(x.to(torch.bfloat16), t).float(),
+            model=lambda x, t, *c: model(
+                x.to(torch.bfloat16), t,
+                null_ctx.expand(x.shape[0], -1, -1).to(torch.bfloat16)).float(),
             noise_schedule=schedule, model_output_transform=transform,
             timestep_spacing="karras")
         if graph:
