@@ -111,7 +111,8 @@ class KarrasPredictionTransform(DiffusionPredictionTransform):
 
     def pred_transform(self, x_t, preds, rates, epsilon=1e-8):
         _, sigma = rates
-        c_out = sigma * self.sigma_data / (torch.sqrt(torch.as_tensor(self.sigma_data ** 2, dtype=sigma.dtype, device=sigma.device) + sigma ** 2) + epsilon)
+        # scalar arithmetic only — graph-capture-safe (no H2D tensor creation)
+        c_out = sigma * self.sigma_data / (torch.sqrt(sigma ** 2 + self.sigma_data ** 2) + epsilon)
         c_skip = self.sigma_data ** 2 / (self.sigma_data ** 2 + sigma ** 2 + epsilon)
         c_out = c_out.reshape(get_coeff_shapes_tuple(preds))
         c_skip = c_skip.reshape(get_coeff_shapes_tuple(x_t))
@@ -119,4 +120,5 @@ class KarrasPredictionTransform(DiffusionPredictionTransform):
 
     def get_input_scale(self, rates, epsilon=1e-8):
         _, sigma = rates
-        return 1 / (torch.sqrt(torch.as_tensor(self.sigma_data ** 2, dtype=sigma.dtype, device=sigma.device) + sigma ** 2) + epsilon)
+        # scalar arithmetic only — graph-capture-safe (no H2D tensor creation)
+        return 1 / (torch.sqrt(sigma ** 2 + self.sigma_data ** 2) + epsilon)
