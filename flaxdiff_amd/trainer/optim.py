@@ -24,7 +24,8 @@ class FlatAdamWEMA:
     def __init__(self, module: torch.nn.Module, lr: float = 2.7e-4,
                  betas=(0.9, 0.999), eps: float = 1e-8, weight_decay: float = 0.0,
                  ema_decay: float = 0.999, lr_schedule: Optional[Callable[[int], float]] = None,
-                 grad_clip_norm: Optional[float] = None):
+                 grad_clip_norm: Optional[float] = None,
+                 skip_nonfinite: bool = True):
         self.lr = lr
         self.beta1, self.beta2 = betas
         self.eps = eps
@@ -32,6 +33,8 @@ class FlatAdamWEMA:
         self.ema_decay = ema_decay
         self.lr_schedule = lr_schedule
         self.grad_clip_norm = grad_clip_norm
+        self.skip_nonfinite = skip_nonfinite
+        self.skipped_steps = 0
         self.step_count = 0
 
         params = [p for p in module.parameters() if p.requires_grad]
@@ -83,11 +86,18 @@ class FlatAdamWEMA:
         self.step_count += 1
         lr = self.current_lr()
 
-        if self.grad_clip_norm is not None:
-            gnorm = self.flat_grad.norm() * grad_scale
-            clip = self.grad_clip_norm / (float(gnorm) + 1e-6)
-            if clip < 1.0:
-                grad_scale = grad_scale * clip
+        if self.grad_clip_norm is not None or self.skip_nonfinite:
+            gnorm = float(self.flat_grad.norm()) * grad_scale
+            if self.skip_nonfinite and not math.isfinite(gnorm):
+                # DynamicScale skip-on-nonfinite parity (reference
+                # diffusion_trainer.py:229-240): drop the step, keep state.
+                self.step_count -= 1
+                self.skipped_steps += 1
+                return
+            if self.grad_clip_norm is not None:
+                clip = self.grad_clip_norm / (gnorm + 1e-6)
+                if clip < 1.0:
+                    grad_scale = grad_scale * clip
 
         if self.flat.is_cuda and ops.hip_available():
             ops.fused_adamw_ema(self.flat, self.flat_grad, self.exp_avg,

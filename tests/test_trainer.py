@@ -123,3 +123,22 @@ def test_grad_clip():
     (m(torch.randn(8, 4)).sum() * 1000).backward()
     opt.step()  # should not blow up; lr=0 keeps params fixed
     assert torch.isfinite(opt.flat).all()
+
+
+def test_nonfinite_grad_skips_step():
+    """DynamicScale parity: a NaN gradient leaves params untouched."""
+    import torch.nn as nn
+    from flaxdiff_amd.trainer.optim import FlatAdamWEMA
+
+    m = nn.Linear(4, 4)
+    opt = FlatAdamWEMA(m, lr=1e-2)
+    before = opt.flat.clone()
+    opt.flat_grad.fill_(float("nan"))
+    opt.step()
+    assert torch.equal(opt.flat, before)
+    assert opt.skipped_steps == 1 and opt.step_count == 0
+    # a finite grad then applies normally
+    opt.flat_grad.fill_(0.1)
+    opt.step()
+    assert not torch.equal(opt.flat, before)
+    assert opt.step_count == 1
