@@ -137,6 +137,11 @@ CONV_SHAPES = [
     (2, 16, 16, 64, 128, 3, 2),      # downsample
     (2, 15, 15, 32, 48, 3, 2),       # odd spatial
     (1, 8, 8, 768, 512, 3, 1),       # decoder concat
+    (2, 64, 64, 64, 64, 3, 1),       # level-0: halo path, 2 rows/tile
+    (2, 8, 8, 512, 512, 3, 1),       # level-3: halo path, 16 rows/tile
+    (1, 32, 32, 128, 128, 3, 1),     # halo path, 4 rows/tile
+    (3, 16, 16, 96, 64, 3, 1),       # halo with batch-crossing tiles
+    (2, 48, 48, 32, 32, 3, 1),       # W=48: 128%48!=0 -> general kernel
 ]
 
 
