@@ -159,7 +159,7 @@ def test_conv2d_fwd(B, H, W, Ci, Co, k, st):
     assert err < 3e-2, f"conv fwd rel err {err}"
 
 
-@pytest.mark.parametrize("B,H,W,Ci,Co,k,st", CONV_SHAPES[:6])
+@pytest.mark.parametrize("B,H,W,Ci,Co,k,st", CONV_SHAPES[:6] + CONV_SHAPES[8:12])
 def test_conv2d_backward(B, H, W, Ci, Co, k, st):
     torch.manual_seed(4)
     x_cpu = torch.randn(B, H, W, Ci) * 0.5
