@@ -76,6 +76,34 @@ class GeneralDiffusionTrainer(DiffusionTrainer):
         return results
 
     # ------------------------------------------------------------------
+    def make_validation_fn(self, sampler_class=None, num_samples: int = 4,
+                           resolution: int = 64, diffusion_steps: int = 50,
+                           guidance_scale: float = 3.0,
+                           conditioning_context=None):
+        """Epoch-end validation hook for fit(): EMA sample generation +
+        eval metrics + wandb logging (reference general_diffusion_trainer
+        :378-519; guidance 3.0 default, :375)."""
+        if sampler_class is None:
+            from ..samplers import EulerAncestralSampler
+            sampler_class = EulerAncestralSampler
+
+        def val_fn(trainer):
+            samples = trainer.validation_sample(
+                sampler_class, num_samples=num_samples, resolution=resolution,
+                diffusion_steps=diffusion_steps, guidance_scale=guidance_scale,
+                conditioning_context=conditioning_context, use_ema=True)
+            batch = {"image": ((samples + 1) * 127.5).clamp(0, 255).byte()}
+            if conditioning_context is not None:
+                batch["text"] = conditioning_context
+            results = trainer.evaluate(samples, batch)
+            if trainer.dist.is_main:
+                trainer.wandb.log({"val/samples_mean": float(samples.mean()),
+                                   **{f"val/{k}": v for k, v in results.items()}})
+            return results
+
+        return val_fn
+
+    # ------------------------------------------------------------------
     def push_to_registry(self, registry_dir: str = "./registry"):
         """Export the best checkpoint locally (stand-in for the reference's
         wandb model-registry push, :560-594)."""

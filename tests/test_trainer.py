@@ -142,3 +142,35 @@ def test_nonfinite_grad_skips_step():
     opt.step()
     assert not torch.equal(opt.flat, before)
     assert opt.step_count == 1
+
+
+def test_validation_fn_end_to_end(tmp_path):
+    """Epoch validation hook: EMA sampling + metrics through fit()."""
+    from flaxdiff_amd.metrics import EvaluationMetric
+    from flaxdiff_amd.models import Unet
+    from flaxdiff_amd.predictors import KarrasPredictionTransform
+    from flaxdiff_amd.schedulers import EDMNoiseScheduler
+    from flaxdiff_amd.trainer import GeneralDiffusionTrainer
+
+    calls = []
+    metric = EvaluationMetric(
+        function=lambda gen, batch: float(gen.float().std()),
+        name="sample_std", higher_is_better=True)
+    model = Unet(emb_features=32, feature_depths=[8, 16],
+                 attention_configs=[None, None], num_res_blocks=1,
+                 num_middle_res_blocks=1, norm_groups=4, context_dim=768)
+    tr = GeneralDiffusionTrainer(
+        model, EDMNoiseScheduler(1, sigma_max=80),
+        KarrasPredictionTransform(sigma_data=0.5),
+        eval_metrics=[metric], name="valfn",
+        checkpoint_base_path=str(tmp_path), distributed=False)
+    val_fn = tr.make_validation_fn(num_samples=2, resolution=16,
+                                   diffusion_steps=2, guidance_scale=0.0)
+
+    def batches():
+        while True:
+            yield {"image": torch.randint(0, 255, (2, 16, 16, 3),
+                                          dtype=torch.uint8)}
+
+    tr.fit(batches(), steps_per_epoch=2, epochs=1, val_fn=val_fn)
+    assert "sample_std" in tr.best_metric_values
