@@ -179,11 +179,15 @@ class _Conv2dFn(torch.autograd.Function):
         dy = dy.contiguous()
         dx = ext.conv2d_dgrad(dy, w, ctx.stride, x.shape[1], x.shape[2]) \
             if ctx.needs_input_grad[0] else None
-        dw = None
+        dw = db = None
         if ctx.needs_input_grad[1]:
-            dw = ext.conv2d_wgrad(dy, x, w.shape[0], w.shape[1], ctx.stride)
+            # wgrad kernel also folds the bias grad (dB accumulated from the
+            # dY tiles it stages anyway)
+            dw, db = ext.conv2d_wgrad(dy, x, w.shape[0], w.shape[1], ctx.stride)
             dw = dw.to(w.dtype)
-        db = dy.reshape(-1, dy.shape[-1]).sum(0).to(w.dtype) if ctx.has_bias else None
+            db = db.to(w.dtype) if ctx.has_bias else None
+        elif ctx.has_bias:
+            db = dy.reshape(-1, dy.shape[-1]).sum(0).to(w.dtype)
         return dx, dw, db, None
 
 
