@@ -338,6 +338,7 @@ def test_sampling_on_gpu():
 
 
 @pytest.mark.gpu
+@pytest.mark.timeout(120, method="thread")
 def test_graph_captured_sampling_matches_eager():
     """hipGraph-replayed sample_model must produce the same samples as eager."""
     from flaxdiff_amd.models import Unet
