@@ -235,10 +235,12 @@ class _AttentionFn(torch.autograd.Function):
     def backward(ctx, do):
         q, k, v, lse = ctx.saved_tensors
         scale = ctx.scale
-        # Small-KV (cross-attention Skv<=128, D<=32): ONE fused hand-written
-        # kernel computes dq/dk/dv with K/V LDS-resident — replaces the
-        # batched tiny-output hipBLASLt GEMMs below (~0.7% MFMA efficiency).
-        if k.shape[2] <= 128 and q.shape[3] <= 32:
+        # Small-KV fused backward: measured SLOWER than the composed GEMM
+        # path at the bench shapes (shfl-latency-bound per-row design,
+        # profiles/r01 b9: 9.4ms/call at Sq=4096) — kept behind an env
+        # opt-in until a fragment-MFMA redesign lands.
+        if (os.environ.get("FLAXDIFF_FUSED_ATTN_BWD", "0") == "1"
+                and k.shape[2] <= 128 and q.shape[3] <= 32):
             ext = _require_ext()
             dq, dk, dv = ext.attn_bwd_smallkv(q, k, v, do.contiguous(), lse,
                                               scale)
