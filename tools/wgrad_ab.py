@@ -1,4 +1,5 @@
 """A/B the wgrad bias fold + kernel timings on the bench shape zoo."""
+import sys, os; sys.path.insert(0, os.getcwd())
 import os, time, json
 import torch
 from flaxdiff_amd.ops import _require_ext
