@@ -105,7 +105,7 @@ class DiffusionInferencePipeline:
                     timestep_spacing: str = "linear") -> DiffusionSampler:
         key = (sampler_class, guidance_scale)
         if key not in self._sampler_cache:
-            self._sampler_cache[key] = sampler_class(
+            sampler = sampler_class(
                 model=self.model,
                 noise_schedule=self.noise_schedule,
                 model_output_transform=self.prediction_transform,
@@ -113,6 +113,10 @@ class DiffusionInferencePipeline:
                 guidance_scale=guidance_scale,
                 autoencoder=self.autoencoder,
                 timestep_spacing=timestep_spacing)
+            # hipGraph-replay the per-step model eval (bit-exact vs eager,
+            # verified in tests/test_gpu_ops.py::test_graph_captured_*)
+            sampler.enable_graph_capture()
+            self._sampler_cache[key] = sampler
         return self._sampler_cache[key]
 
     # ------------------------------------------------------------------
