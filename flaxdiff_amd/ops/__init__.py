@@ -181,12 +181,10 @@ class _Conv2dFn(torch.autograd.Function):
             if ctx.needs_input_grad[0] else None
         dw = db = None
         if ctx.needs_input_grad[1]:
-            # wgrad kernel also folds the bias grad (dB accumulated from the
-            # dY tiles it stages anyway)
-            dw, db = ext.conv2d_wgrad(dy, x, w.shape[0], w.shape[1], ctx.stride)
+            dw, _db_unused = ext.conv2d_wgrad(dy, x, w.shape[0], w.shape[1],
+                                              ctx.stride)
             dw = dw.to(w.dtype)
-            db = db.to(w.dtype) if ctx.has_bias else None
-        elif ctx.has_bias:
+        if ctx.has_bias:
             db = dy.reshape(-1, dy.shape[-1]).sum(0).to(w.dtype)
         return dx, dw, db, None
 
