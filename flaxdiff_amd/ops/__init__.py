@@ -233,12 +233,10 @@ class _AttentionFn(torch.autograd.Function):
     def backward(ctx, do):
         q, k, v, lse = ctx.saved_tensors
         scale = ctx.scale
-        # Small-KV fused backward: measured SLOWER than the composed GEMM
-        # path at the bench shapes (shfl-latency-bound per-row design,
-        # profiles/r01 b9: 9.4ms/call at Sq=4096) — kept behind an env
-        # opt-in until a fragment-MFMA redesign lands.
-        if (os.environ.get("FLAXDIFF_FUSED_ATTN_BWD", "0") == "1"
-                and k.shape[2] <= 128 and q.shape[3] <= 32):
+        # Small-KV fused MFMA flash backward: 3x the composed GEMM path at
+        # the bench shapes (2.34 vs 7.12 ms at Sq=4096 D=16 — tools/
+        # attn_bwd_ab.py). Larger head dims fall through to the GEMMs.
+        if k.shape[2] <= 128 and q.shape[3] <= 32:
             ext = _require_ext()
             dq, dk, dv = ext.attn_bwd_smallkv(q, k, v, do.contiguous(), lse,
                                               scale)
