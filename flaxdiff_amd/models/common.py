@@ -91,10 +91,8 @@ class Dense(nn.Module):
         self.bias = nn.Parameter(torch.zeros(out_features)) if use_bias else None
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        y = torch.matmul(x, _cast(self.weight, x.dtype))
-        if self.bias is not None:
-            y = y + _cast(self.bias, x.dtype)
-        return y
+        return ops.dense(x, _cast(self.weight, x.dtype),
+                         _cast(self.bias, x.dtype))
 
 
 # ---------------------------------------------------------------------------
