@@ -92,8 +92,12 @@ class S5Layer(nn.Module):
         C_c = torch.complex(self.C_re.float(), self.C_im.float())  # [F, N]
 
         Bu = torch.einsum("bsf,nf->bsn", u.to(B_bar.dtype), B_bar)  # [B,S,N] cplx
-        a = A_bar.reshape(1, 1, -1).expand(B, S, self.state_dim)
-        x_states = associative_scan_diag(a, Bu)
+        if Bu.is_cuda:
+            from .. import ops
+            x_states = ops.s5_scan(A_bar, Bu)
+        else:
+            a = A_bar.reshape(1, 1, -1).expand(B, S, self.state_dim)
+            x_states = associative_scan_diag(a, Bu)
 
         y = torch.einsum("fn,bsn->bsf", C_c, x_states).real
         y = y + self.D.float().reshape(1, 1, -1) * u

@@ -357,6 +357,37 @@ def fourier_time_embedding(t: torch.Tensor, freqs: torch.Tensor) -> torch.Tensor
 # Fused Adam + EMA optimizer step (HIP kernel, one pass over params)
 # ---------------------------------------------------------------------------
 
+class _S5ScanFn(torch.autograd.Function):
+    """x_s = a * x_{s-1} + bu_s (complex, a constant per channel) — serial
+    per-(b,n) HIP scan (one read + one write of the tensor) replacing the
+    log-depth doubling. Backward is the exact reverse scan with conj(a)."""
+
+    @staticmethod
+    def forward(ctx, a_ri, bu_ri):
+        ext = _require_ext()
+        x = ext.s5_scan_fwd(a_ri.contiguous(), bu_ri.contiguous())
+        ctx.save_for_backward(a_ri, x)
+        return x
+
+    @staticmethod
+    def backward(ctx, dx):
+        a_ri, x = ctx.saved_tensors
+        ext = _require_ext()
+        dbu, dap = ext.s5_scan_bwd(a_ri.contiguous(), x, dx.contiguous())
+        return dap.sum(dim=0), dbu
+
+
+def s5_scan(a: torch.Tensor, bu: torch.Tensor) -> torch.Tensor:
+    """a: [N] complex64; bu: [B,S,N] complex64 -> inclusive scan [B,S,N]."""
+    if bu.is_cuda and not _FORCE_TORCH:
+        a_ri = torch.view_as_real(a.contiguous()).float()
+        bu_ri = torch.view_as_real(bu.contiguous()).float()
+        x = _S5ScanFn.apply(a_ri, bu_ri)
+        return torch.view_as_complex(x)
+    from ..models.ssm_dit import associative_scan_diag
+    return associative_scan_diag(a.reshape(1, 1, -1).expand_as(bu), bu)
+
+
 def fused_adamw_ema(params_f32, grads, exp_avg, exp_avg_sq, ema_f32, params_bf16,
                     *, lr, beta1, beta2, eps, weight_decay, step, ema_decay,
                     grad_scale: float = 1.0):
