@@ -346,10 +346,16 @@ def avg_pool_2x(x: torch.Tensor) -> torch.Tensor:
 
 def sinusoidal_time_embedding(t: torch.Tensor, features: int,
                               max_positions: int = 10000) -> torch.Tensor:
+    if _use_hip(t):
+        ext = _require_ext()
+        return ext.time_embed(t, torch.Tensor(), features, float(max_positions))
     return reference.sinusoidal_time_embedding(t, features, max_positions)
 
 
 def fourier_time_embedding(t: torch.Tensor, freqs: torch.Tensor) -> torch.Tensor:
+    if _use_hip(t):
+        ext = _require_ext()
+        return ext.time_embed(t, freqs, freqs.numel() * 2, 0.0)
     return reference.fourier_time_embedding(t, freqs)
 
 

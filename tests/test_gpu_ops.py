@@ -518,3 +518,16 @@ def test_hybrid_ssm_dit_gpu_train_step():
     batch = {"image": torch.randint(0, 255, (2, 32, 32, 3), dtype=torch.uint8)}
     out = tr.train_step(batch)
     assert out["loss"] == out["loss"]
+
+
+@pytest.mark.gpu
+def test_time_embed_kernel_matches_reference():
+    t = torch.rand(16, device="cuda") * 1000
+    ref = reference.sinusoidal_time_embedding(t.cpu(), 64)
+    got = ops.sinusoidal_time_embedding(t, 64)
+    assert rel_err(got.cpu(), ref) < 1e-4
+    g = torch.Generator().manual_seed(42)
+    freqs = (torch.randn(32, generator=g) * 16).cuda()
+    ref = reference.fourier_time_embedding(t.cpu(), freqs.cpu())
+    got = ops.fourier_time_embedding(t, freqs)
+    assert rel_err(got.cpu(), ref) < 1e-4
