@@ -185,7 +185,7 @@ class _Conv2dFn(torch.autograd.Function):
                                               ctx.stride)
             dw = dw.to(w.dtype)
         if ctx.has_bias:
-            db = dy.reshape(-1, dy.shape[-1]).sum(0).to(w.dtype)
+            db = _bias_grad(dy).to(w.dtype)
         return dx, dw, db, None
 
 
@@ -217,7 +217,7 @@ class _DenseFn(torch.autograd.Function):
             dw4, _ = ext.conv2d_wgrad(dy.view(M, 1, 1, dy.shape[1]),
                                       x2d.view(M, 1, 1, x2d.shape[1]), 1, 1, 1)
             dw = dw4.view(x2d.shape[1], dy.shape[1]).to(w2d.dtype)
-        db = dy.sum(0) if ctx.has_bias else None
+        db = _bias_grad(dy).to(dy.dtype) if ctx.has_bias else None
         return dx, dw, db
 
 
@@ -233,6 +233,15 @@ def dense(x: torch.Tensor, w: torch.Tensor, b: Optional[torch.Tensor] = None
     if b is not None:
         y = y + b
     return y
+
+
+def _bias_grad(dy: torch.Tensor) -> torch.Tensor:
+    """Column sum over all but the channel dim (coalesced HIP kernel)."""
+    d2 = dy.reshape(-1, dy.shape[-1])
+    C = d2.shape[1]
+    if d2.is_cuda and d2.dtype == torch.bfloat16 and C % 8 == 0 and C <= 2048:
+        return _require_ext().colsum(d2.contiguous())
+    return d2.float().sum(0)
 
 
 def conv2d(x: torch.Tensor, w: torch.Tensor, b: Optional[torch.Tensor] = None,

@@ -553,3 +553,13 @@ def test_dense_wgrad_splitm_matches_autograd():
     assert rel_err(x.grad, xf.grad) < 5e-2
     assert rel_err(w.grad, wf.grad) < 5e-2
     assert rel_err(b.grad, dy.float().sum(0)) < 5e-2
+
+
+@pytest.mark.gpu
+def test_colsum_matches_torch():
+    from flaxdiff_amd.ops import _require_ext
+    torch.manual_seed(0)
+    x = (torch.randn(10000, 96, device="cuda")).bfloat16()
+    got = _require_ext().colsum(x)
+    want = x.float().sum(0)
+    assert rel_err(got, want) < 1e-2
