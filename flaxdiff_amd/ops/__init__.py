@@ -260,6 +260,18 @@ def conv2d(x: torch.Tensor, w: torch.Tensor, b: Optional[torch.Tensor] = None,
 
 def conv2d_transpose(x: torch.Tensor, w: torch.Tensor, b: Optional[torch.Tensor] = None,
                      stride: int = 2) -> torch.Tensor:
+    """flax-SAME transposed conv. On GPU: zero-stuff + spatially-flipped
+    SAME conv through the native implicit-GEMM kernel (convT(x) ==
+    conv(pad(zerostuff(x)), flip(w)) cropped to H*stride)."""
+    kh, kw = w.shape[0], w.shape[1]
+    if _use_hip(x) and stride == 2 and kh == 3 and kw == 3:
+        B, H, W, Ci = x.shape
+        stuffed = torch.zeros(B, 2 * H + 1, 2 * W + 1, Ci, device=x.device,
+                              dtype=x.dtype)
+        stuffed[:, 1::2, 1::2] = x
+        wf = torch.flip(w, dims=(0, 1)).contiguous()
+        out = conv2d(stuffed, wf, b, stride=1)
+        return out[:, : 2 * H, : 2 * W].contiguous()
     return reference.conv2d_transpose_nhwc(x, w, b, stride)
 
 

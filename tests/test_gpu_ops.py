@@ -600,3 +600,19 @@ def test_pipeline_cfg_graph_sampling(tmp_path):
     # graph was actually captured for the CFG-doubled batch
     sampler = pipe.get_sampler(EulerAncestralSampler, 2.0)
     assert sampler._graphed is not None
+
+
+@pytest.mark.gpu
+def test_conv_transpose_gpu_matches_reference():
+    torch.manual_seed(0)
+    x = torch.randn(2, 8, 8, 16) * 0.5
+    w = torch.randn(3, 3, 16, 24) * 0.1
+    b = torch.randn(24) * 0.1
+    ref = reference.conv2d_transpose_nhwc(x, w, b, 2)
+    xg = x.bfloat16().cuda().requires_grad_(True)
+    wg = w.bfloat16().cuda().requires_grad_(True)
+    y = ops.conv2d_transpose(xg, wg, b.bfloat16().cuda(), 2)
+    assert y.shape == ref.shape
+    assert rel_err(y.cpu(), ref) < 4e-2
+    y.float().sum().backward()
+    assert torch.isfinite(xg.grad.float()).all()
