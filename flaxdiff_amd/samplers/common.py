@@ -47,6 +47,7 @@ class DiffusionSampler:
 
         self._graph_capture = False
         self._graphed: Optional["_GraphedSampleModel"] = None
+        self._uncond_cache = {}
 
     def enable_graph_capture(self, enabled: bool = True):
         """Capture the per-step CFG-doubled model forward in a hipGraph and
@@ -73,10 +74,18 @@ class DiffusionSampler:
             if torch.is_tensor(c_in):
                 c_in = c_in.to(x_cat.dtype)
             finals = []
-            for cond, uncond in zip(conditioning_inputs, self.unconditionals):
-                uncond = torch.as_tensor(uncond, device=cond.device, dtype=cond.dtype)
-                uncond = uncond.broadcast_to(cond.shape)
-                finals.append(torch.cat([cond, uncond], dim=0))
+            for i, (cond, uncond) in enumerate(zip(conditioning_inputs,
+                                                   self.unconditionals)):
+                # cache the null embedding device-resident: an H2D copy here
+                # would be illegal inside hipGraph capture
+                key = (i, cond.device, cond.dtype)
+                uncond_dev = self._uncond_cache.get(key)
+                if uncond_dev is None:
+                    uncond_dev = torch.as_tensor(uncond, device=cond.device,
+                                                 dtype=cond.dtype)
+                    self._uncond_cache[key] = uncond_dev
+                finals.append(torch.cat(
+                    [cond, uncond_dev.broadcast_to(cond.shape)], dim=0))
             xin, tin = ns.transform_inputs(x_cat * c_in, t_cat)
             out = self.model(xin, tin.to(x_cat.device) if torch.is_tensor(tin) else tin, *finals)
             out_cond, out_uncond = out.chunk(2, dim=0)
