@@ -33,6 +33,11 @@ class DiffusionInferencePipeline:
         self.step = step
         self._sampler_cache: Dict[Tuple[type, float], DiffusionSampler] = {}
         self._param_source = "params"
+        # GPU inference runs the model in bf16 (the HIP kernel set is bf16;
+        # sampler-side state stays fp32)
+        self.compute_dtype = (torch.bfloat16 if torch.cuda.is_available()
+                              else torch.float32)
+        self.device = "cuda" if torch.cuda.is_available() else "cpu"
 
     # ------------------------------------------------------------------
     @classmethod
@@ -47,6 +52,7 @@ class DiffusionInferencePipeline:
                    parts["raw_config"], parts["step"])
         pipe.load_params(use_ema=use_ema, use_best=use_best)
         dev = device or ("cuda" if torch.cuda.is_available() else "cpu")
+        pipe.device = dev
         pipe.model.to(dev).eval()
         return pipe
 
@@ -105,8 +111,15 @@ class DiffusionInferencePipeline:
                     timestep_spacing: str = "linear") -> DiffusionSampler:
         key = (sampler_class, guidance_scale)
         if key not in self._sampler_cache:
+            dt = self.compute_dtype
+            model = self.model
+            if dt != torch.float32:
+                def model_fn(x, t, *conds, _m=self.model, _dt=dt):
+                    return _m(x.to(_dt), t,
+                              *(c.to(_dt) for c in conds)).float()
+                model = model_fn
             sampler = sampler_class(
-                model=self.model,
+                model=model,
                 noise_schedule=self.noise_schedule,
                 model_output_transform=self.prediction_transform,
                 input_config=self.input_config,
@@ -137,4 +150,4 @@ class DiffusionInferencePipeline:
             num_samples=num_samples, resolution=resolution,
             sequence_length=sequence_length, diffusion_steps=diffusion_steps,
             start_step=start_step, conditioning=conditioning,
-            rngstate=rngstate, progress=progress)
+            device=self.device, rngstate=rngstate, progress=progress)
