@@ -2,7 +2,8 @@ from .sources import (AUGMENT_MODE_ENV, DataAugmenter, DataSource,
                       HFDatasetSource, ImageAugmenter, ImageFolderSource,
                       SyntheticImageSource, TensorSource, datasetMap,
                       register_dataset, register_image_folder)
-from .dataloaders import (AugmentedDataset, PrefetchLoader, ShardedSampler,
+from .dataloaders import (AugmentedDataset, DevicePrefetcher, PrefetchLoader,
+                          ShardedSampler,
                           collate_image_batch, get_dataset, get_dataset_online,
                           make_dataloader)
 
@@ -10,7 +11,7 @@ __all__ = [
     "AUGMENT_MODE_ENV", "DataAugmenter", "DataSource", "HFDatasetSource",
     "ImageAugmenter", "ImageFolderSource", "SyntheticImageSource",
     "TensorSource", "datasetMap", "register_dataset", "register_image_folder",
-    "AugmentedDataset", "PrefetchLoader", "ShardedSampler",
+    "AugmentedDataset", "DevicePrefetcher", "PrefetchLoader", "ShardedSampler",
     "collate_image_batch", "get_dataset", "get_dataset_online",
     "make_dataloader",
 ]

@@ -207,3 +207,50 @@ class PrefetchLoader:
 
     def stop(self):
         self._stop.set()
+
+
+class DevicePrefetcher:
+    """Wraps a host loader and keeps the NEXT batch's H2D copies in flight on
+    a side HIP stream while the current step computes (the reference's
+    DataLoaderWithMesh pushed device arrays from a background thread,
+    dataloaders.py:28-82)."""
+
+    def __init__(self, loader, device, compute_dtype=None):
+        self.loader = loader
+        self.device = torch.device(device)
+        self.compute_dtype = compute_dtype
+        self._use_stream = self.device.type == "cuda"
+        self.stream = torch.cuda.Stream() if self._use_stream else None
+
+    def _to_device(self, batch):
+        out = {}
+        for k, v in batch.items():
+            if torch.is_tensor(v):
+                out[k] = v.to(self.device, non_blocking=True)
+            elif isinstance(v, dict):
+                out[k] = {kk: vv.to(self.device, non_blocking=True)
+                          if torch.is_tensor(vv) else vv
+                          for kk, vv in v.items()}
+            else:
+                out[k] = v
+        return out
+
+    def __iter__(self):
+        it = iter(self.loader)
+        if not self._use_stream:
+            for batch in it:
+                yield self._to_device(batch)
+            return
+        nxt = None
+        for batch in it:
+            with torch.cuda.stream(self.stream):
+                staged = self._to_device(batch)
+            if nxt is not None:
+                yield nxt
+            torch.cuda.current_stream().wait_stream(self.stream)
+            nxt = staged
+        if nxt is not None:
+            yield nxt
+
+    def __len__(self):
+        return len(self.loader)

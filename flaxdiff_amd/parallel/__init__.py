@@ -73,6 +73,17 @@ def init_distributed(backend: Optional[str] = None) -> DistContext:
     return DistContext(rank=rank, world_size=world, local_rank=local_rank, device=device)
 
 
+_RANK_COLORS = [32, 33, 34, 35, 36, 31, 92, 93]  # ANSI per-rank
+
+
+def rank_print(*args, rank: Optional[int] = None, **kwargs):
+    """Rank-colored console logging (reference simple_trainer.py:32-41)."""
+    if rank is None:
+        rank = dist.get_rank() if dist.is_initialized() else 0
+    color = _RANK_COLORS[rank % len(_RANK_COLORS)]
+    print(f"\033[{color}m[rank {rank}]\033[0m", *args, **kwargs)
+
+
 def barrier():
     if dist.is_initialized():
         dist.barrier()
