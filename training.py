@@ -312,12 +312,16 @@ def main(argv=None):
 
     steps_per_epoch = args.steps_per_epoch or max(len(loader), 1)
 
+    # H2D copies of the next batch overlap the current step on a side stream
+    from flaxdiff_amd.data import DevicePrefetcher
+    prefetched = DevicePrefetcher(loader, trainer.device)
+
     def data_iter():
         epoch = 0
         while True:
             if hasattr(loader, "sampler") and hasattr(loader.sampler, "set_epoch"):
                 loader.sampler.set_epoch(epoch)
-            for batch in loader:
+            for batch in prefetched:
                 yield batch
             epoch += 1
 
