@@ -98,3 +98,56 @@ def test_per_rank_rng_decorrelated():
             p.join(timeout=100)
         res = dict(results)
     assert res[0] != res[1]
+
+
+def _cfg_worker(rank, world, port, results):
+    os.environ.update({
+        "RANK": str(rank), "WORLD_SIZE": str(world), "LOCAL_RANK": str(rank),
+        "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+    })
+    import torch.distributed as dist
+    from flaxdiff_amd.predictors import KarrasPredictionTransform
+    from flaxdiff_amd.schedulers import EDMNoiseScheduler
+    from flaxdiff_amd.trainer import DiffusionTrainer
+
+    torch.manual_seed(7)
+    m = Unet(emb_features=32, feature_depths=[8, 16],
+             attention_configs=[{"heads": 2}, {"heads": 2}], num_res_blocks=1,
+             num_middle_res_blocks=1, norm_groups=4, context_dim=16)
+    tr = DiffusionTrainer(m, EDMNoiseScheduler(1, sigma_max=80),
+                          KarrasPredictionTransform(sigma_data=0.5),
+                          name="ddp-cfg", checkpoint_base_path="/tmp/fd_ddp_cfg",
+                          text_context_shape=(4, 16), unconditional_prob=0.5,
+                          distributed=True)
+    g = torch.Generator().manual_seed(300 + rank)
+    batch = {"image": torch.randint(0, 255, (4, 16, 16, 3), generator=g,
+                                    dtype=torch.uint8),
+             "text_emb": torch.randn(4, 4, 16, generator=g)}
+    losses = [tr.train_step(batch)["loss"] for _ in range(2)]
+    flat = tr.optimizer.flat.clone()
+    gathered = [torch.zeros_like(flat) for _ in range(world)]
+    dist.all_gather(gathered, flat)
+    same = all(torch.allclose(gathered[0], gi, atol=1e-6) for gi in gathered)
+    results[rank] = {"losses": losses, "params_equal": bool(same)}
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_data_parallel_text_conditional_cfg():
+    """Text-conditional + CFG-dropout training stays in sync across ranks
+    (BASELINE config 4's DP path, EDM schedule)."""
+    world = 2
+    ctx = mp.get_context("spawn")
+    with ctx.Manager() as man:
+        results = man.dict()
+        procs = [ctx.Process(target=_cfg_worker, args=(r, world, 29733, results))
+                 for r in range(world)]
+        for p in procs:
+            p.start()
+        for p in procs:
+            p.join(timeout=240)
+        for p in procs:
+            assert p.exitcode == 0, f"worker failed: {p.exitcode}"
+        res = dict(results)
+    assert res[0]["params_equal"] and res[1]["params_equal"]
+    assert res[0]["losses"] == pytest.approx(res[1]["losses"], rel=1e-5)
