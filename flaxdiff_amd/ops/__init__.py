@@ -275,9 +275,39 @@ def conv2d_transpose(x: torch.Tensor, w: torch.Tensor, b: Optional[torch.Tensor]
     return reference.conv2d_transpose_nhwc(x, w, b, stride)
 
 
+class _DepthwiseConvFn(torch.autograd.Function):
+    """3x3 depthwise NHWC conv on the HIP kernels (depthwise.hip)."""
+
+    @staticmethod
+    def forward(ctx, x, w, stride, dilation):
+        ext = _require_ext()
+        y = ext.dwconv_fwd(x, w, None, stride, dilation)
+        ctx.save_for_backward(x, w)
+        ctx.stride = stride
+        ctx.dilation = dilation
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w = ctx.saved_tensors
+        ext = _require_ext()
+        dy = dy.contiguous()
+        dx = ext.dwconv_dgrad(dy, w, x.shape[1], x.shape[2],
+                              ctx.stride, ctx.dilation)
+        dw = ext.dwconv_wgrad(dy, x, ctx.stride, ctx.dilation)
+        return dx, dw.to(w.dtype), None, None
+
+
 def depthwise_conv2d(x: torch.Tensor, w: torch.Tensor, b: Optional[torch.Tensor] = None,
-                     stride: int = 1) -> torch.Tensor:
-    return reference.depthwise_conv2d_nhwc(x, w, b, stride)
+                     stride: int = 1, dilation: int = 1) -> torch.Tensor:
+    """Depthwise NHWC conv; w is [kh, kw, C, 1] (SeparableConv,
+    reference common.py:126-153). GPU bf16 3x3 runs the hand-written
+    kernels; other kernel sizes compose torch ops (cold path)."""
+    if (_use_hip(x) and x.dtype == torch.bfloat16 and b is None
+            and w.shape[0] == 3 and w.shape[1] == 3 and x.shape[-1] % 8 == 0):
+        w3 = w.reshape(3, 3, -1).contiguous()
+        return _DepthwiseConvFn.apply(x, w3, stride, dilation)
+    return reference.depthwise_conv2d_nhwc(x, w, b, stride, dilation)
 
 
 # ---------------------------------------------------------------------------

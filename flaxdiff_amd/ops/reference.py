@@ -71,14 +71,16 @@ def conv2d_transpose_nhwc(x: torch.Tensor, w: torch.Tensor, b: torch.Tensor | No
 
 
 def depthwise_conv2d_nhwc(x: torch.Tensor, w: torch.Tensor, b: torch.Tensor | None = None,
-                          stride: int = 1) -> torch.Tensor:
-    """Depthwise NHWC conv; w is [kh, kw, C, 1] (SeparableConv, common.py:126-153)."""
-    kh, kw, c, _ = w.shape
+                          stride: int = 1, dilation: int = 1) -> torch.Tensor:
+    """Depthwise NHWC conv; w is [kh, kw, C, 1] (SeparableConv, common.py:126-153).
+    Symmetric torch-style padding ((k-1)*dilation)//2 per side."""
+    kh, kw, c = w.shape[0], w.shape[1], w.shape[2]
     xc = x.permute(0, 3, 1, 2).contiguous()
-    wc = w.permute(2, 3, 0, 1).contiguous()  # (C,1,kh,kw)
-    pt = (kh - 1) // 2
-    xc = F.pad(xc, (pt, kh - 1 - pt, pt, kh - 1 - pt))
-    out = F.conv2d(xc, wc, b, stride=stride, groups=c)
+    wc = w.reshape(kh, kw, c).permute(2, 0, 1).unsqueeze(1).contiguous()  # (C,1,kh,kw)
+    pt = ((kh - 1) * dilation) // 2
+    pl = ((kw - 1) * dilation) // 2
+    xc = F.pad(xc, (pl, (kw - 1) * dilation - pl, pt, (kh - 1) * dilation - pt))
+    out = F.conv2d(xc, wc, b, stride=stride, groups=c, dilation=dilation)
     return out.permute(0, 2, 3, 1).contiguous()
 
 
