@@ -69,6 +69,22 @@ class DiffusionInferencePipeline:
         path = artifact.download()
         return cls.from_checkpoint(path, **kwargs)
 
+    @classmethod
+    def from_wandb_registry(cls, modelname: str, project: str,
+                            entity: Optional[str] = None,
+                            version: str = "latest",
+                            registry: str = "wandb-registry-model",
+                            **kwargs) -> "DiffusionInferencePipeline":
+        """Model-registry restore (reference pipeline.py:104-143): fetch the
+        named registry artifact, then defer to from_checkpoint."""
+        import wandb
+        api = wandb.Api()
+        prefix = f"{entity}/" if entity else ""
+        artifact = api.artifact(
+            f"{prefix}{registry}/{modelname}:{version}", type="model")
+        path = artifact.download()
+        return cls.from_checkpoint(path, **kwargs)
+
     # ------------------------------------------------------------------
     def load_params(self, use_ema: bool = True, use_best: bool = False):
         """Select params/ema_params (optionally from best_state) and load them
