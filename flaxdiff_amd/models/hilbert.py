@@ -153,3 +153,55 @@ def hilbert_unpatchify(x: torch.Tensor, inv_idx: torch.Tensor, patch_size: int,
 
 
 zigzag_unpatchify = hilbert_unpatchify
+
+
+# ---- visualization helpers (reference hilbert.py:373+) ---------------------
+
+def create_patch_grid(patches_np: np.ndarray, patch_size: int, channels: int,
+                      grid_cols: int = 10, border: int = 1) -> np.ndarray:
+    """Tile a [N, P*P*C] patch sequence into one grid image (numpy), with a
+    border between cells — patches appear in sequence order, which makes the
+    scan order visible."""
+    n = patches_np.shape[0]
+    rows = (n + grid_cols - 1) // grid_cols
+    cell = patch_size + border
+    grid = np.ones((rows * cell + border, grid_cols * cell + border, channels),
+                   dtype=np.float32) * 0.5
+    for k in range(n):
+        r, c = divmod(k, grid_cols)
+        patch = patches_np[k].reshape(patch_size, patch_size, channels)
+        y = border + r * cell
+        x = border + c * cell
+        grid[y:y + patch_size, x:x + patch_size] = patch
+    return grid
+
+
+def visualize_hilbert_curve(h: int, w: int, patch_size: int = 1):
+    """Plot the Hilbert traversal over an h x w grid (requires matplotlib)."""
+    import matplotlib.pyplot as plt  # optional dep; not in the base image
+    idx = _hilbert_indices_np(h // patch_size, w // patch_size)
+    w_p = w // patch_size
+    ys, xs = np.divmod(idx, w_p)
+    fig, ax = plt.subplots(figsize=(6, 6))
+    ax.plot(xs + 0.5, ys + 0.5, "-o", markersize=3)
+    ax.set_xlim(0, w_p)
+    ax.set_ylim(w_p, 0)
+    ax.set_title(f"Hilbert curve {h//patch_size}x{w//patch_size}")
+    ax.set_aspect("equal")
+    return fig
+
+
+def demo_hilbert_patching(image: np.ndarray, patch_size: int = 8):
+    """Show an image next to its Hilbert-ordered patch grid (matplotlib)."""
+    import matplotlib.pyplot as plt  # optional dep
+    x = torch.from_numpy(np.asarray(image, dtype=np.float32))[None]
+    patches, inv = hilbert_patchify(x, patch_size)
+    grid = create_patch_grid(patches[0].numpy(), patch_size, x.shape[-1])
+    fig, axes = plt.subplots(1, 2, figsize=(12, 6))
+    axes[0].imshow(np.asarray(image))
+    axes[0].set_title("input")
+    axes[1].imshow(grid.squeeze())
+    axes[1].set_title("patches in Hilbert order")
+    for a in axes:
+        a.axis("off")
+    return fig

@@ -91,3 +91,83 @@ def clip_images(images: torch.Tensor, clip_min: float = -1.0, clip_max: float = 
 def get_coeff_shapes_tuple(array: torch.Tensor):
     """Reference: schedulers/common.py:6-8 — broadcast shape (-1, 1, 1, ...)."""
     return (-1,) + (1,) * (array.ndim - 1)
+
+
+def denormalize_images(images: torch.Tensor, target_type=torch.uint8,
+                       source_range=(-1.0, 1.0), target_range=(0, 255)) -> torch.Tensor:
+    """Map images from a normalized range to a display range
+    (reference: flaxdiff/utils.py:114-135)."""
+    s_lo, s_hi = source_range
+    t_lo, t_hi = target_range
+    x = (images - s_lo) / (s_hi - s_lo)
+    x = x * (t_hi - t_lo) + t_lo
+    if target_type in (torch.uint8, torch.int32, torch.int64):
+        x = x.round().clamp(t_lo, t_hi)
+    return x.to(target_type)
+
+
+def normalize_images(images: torch.Tensor) -> torch.Tensor:
+    """uint8 [0,255] -> fp32 [-1,1] (the trainer's input normalization,
+    reference diffusion_trainer.py:171)."""
+    return images.float() / 127.5 - 1.0
+
+
+def get_latest_checkpoint(checkpoint_path: str) -> str:
+    """Latest step subdirectory of a checkpoint dir
+    (reference: flaxdiff/utils.py:84-91)."""
+    import os
+    steps = sorted(int(d) for d in os.listdir(checkpoint_path)
+                   if d.isdigit())
+    if not steps:
+        raise FileNotFoundError(f"no checkpoints under {checkpoint_path}")
+    return os.path.join(checkpoint_path, str(steps[-1]))
+
+
+def serialize_model(model) -> dict:
+    """JSON-safe dict of a module's constructor config
+    (reference: flaxdiff/utils.py:60-82). Uses the module's `config` attr if
+    present, else its non-private, non-tensor attributes."""
+    def _clean(v):
+        if isinstance(v, dict):
+            return {k: _clean(x) for k, x in v.items()}
+        if isinstance(v, (list, tuple)):
+            return [_clean(x) for x in v]
+        if isinstance(v, (str, int, float, bool)) or v is None:
+            return v
+        if callable(v):
+            return getattr(v, "__name__", str(v))
+        return str(v)
+    if hasattr(model, "config") and isinstance(model.config, dict):
+        return _clean(model.config)
+    src = {k: v for k, v in vars(model).items()
+           if not k.startswith("_") and not torch.is_tensor(v)}
+    return _clean(src)
+
+
+class AutoTextTokenizer:
+    """CLIP tokenizer to the trainer's batch contract
+    (reference: flaxdiff/utils.py:239-258)."""
+
+    def __init__(self, tensor_type: str = "pt",
+                 modelname: str = "openai/clip-vit-large-patch14"):
+        from transformers import AutoTokenizer
+        self.tokenizer = AutoTokenizer.from_pretrained(modelname)
+        self.tensor_type = tensor_type
+
+    def __call__(self, inputs):
+        tokens = self.tokenizer(
+            inputs, padding="max_length",
+            max_length=self.tokenizer.model_max_length, truncation=True,
+            return_tensors=self.tensor_type)
+        return {"input_ids": tokens["input_ids"],
+                "attention_mask": tokens["attention_mask"],
+                "caption": inputs}
+
+    def __repr__(self):
+        return self.__class__.__name__ + "()"
+
+
+def defaultTextEncodeModel(modelname: str = "openai/clip-vit-large-patch14"):
+    """Default CLIP text encoder (reference: flaxdiff/utils.py:261-263)."""
+    from ..inputs.encoders import CLIPTextEncoder
+    return CLIPTextEncoder.from_modelname(modelname=modelname)
