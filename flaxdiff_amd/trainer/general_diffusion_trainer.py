@@ -19,6 +19,19 @@ from ..utils import get_coeff_shapes_tuple
 from .diffusion_trainer import DiffusionTrainer
 
 
+def generate_modelname(dataset_name: str, noise_schedule_name: str = "",
+                       architecture_name: str = "", model=None,
+                       input_config=None, autoencoder=None,
+                       frames_per_sample: Optional[int] = None) -> str:
+    """Canonical run name (reference general_diffusion_trainer.py:32-106 —
+    its richer template is commented out upstream; the live format is
+    `diffusion-{dataset}-res{H}`)."""
+    res = ""
+    if input_config is not None:
+        res = f"-res{input_config.sample_data_shape[-2]}"
+    return f"diffusion-{dataset_name}{res}"
+
+
 class GeneralDiffusionTrainer(DiffusionTrainer):
     def __init__(self, model, noise_schedule, model_output_transform=None, *,
                  input_config=None, eval_metrics: Optional[List] = None, **kwargs):
