@@ -2,7 +2,9 @@ from .sources import (AUGMENT_MODE_ENV, DataAugmenter, DataSource,
                       HFDatasetSource, ImageAugmenter, ImageFolderSource,
                       SyntheticImageSource, TensorSource, datasetMap,
                       register_dataset, register_image_folder)
-from .dataloaders import (AugmentedDataset, DevicePrefetcher, PrefetchLoader,
+from .dataloaders import (AugmentedDataset, CaptionDeletionTransform,
+                          DevicePrefetcher, PrefetchLoader,
+                          generate_collate_fn,
                           ShardedSampler,
                           collate_image_batch, get_dataset, get_dataset_online,
                           make_dataloader)
@@ -11,7 +13,8 @@ __all__ = [
     "AUGMENT_MODE_ENV", "DataAugmenter", "DataSource", "HFDatasetSource",
     "ImageAugmenter", "ImageFolderSource", "SyntheticImageSource",
     "TensorSource", "datasetMap", "register_dataset", "register_image_folder",
-    "AugmentedDataset", "DevicePrefetcher", "PrefetchLoader", "ShardedSampler",
+    "AugmentedDataset", "CaptionDeletionTransform", "DevicePrefetcher",
+    "PrefetchLoader", "ShardedSampler", "generate_collate_fn",
     "collate_image_batch", "get_dataset", "get_dataset_online",
     "make_dataloader",
 ]

@@ -79,6 +79,36 @@ def collate_image_batch(samples: List[Dict[str, Any]]) -> Dict[str, Any]:
     return batch
 
 
+class CaptionDeletionTransform:
+    """Drop the caption key from samples — unconditional training at the data
+    level (reference dataloaders.py CaptionDeletionTransform)."""
+
+    def __call__(self, element: Dict[str, Any]) -> Dict[str, Any]:
+        element.pop("caption", None)
+        return element
+
+    map = __call__  # grain-style MapTransform interface
+
+
+def generate_collate_fn(media_type: str = "image", tokenizer=None):
+    """Collate factory keyed by media type (reference dataloaders.py
+    generate_collate_fn): tokenizes raw captions when a tokenizer is given."""
+    from .videos import collate_video_batch
+
+    base = collate_image_batch if media_type == "image" else collate_video_batch
+
+    def collate(samples: List[Dict[str, Any]]) -> Dict[str, Any]:
+        batch = base(samples)
+        if tokenizer is not None and "text" not in batch:
+            captions = [s.get("caption", "") for s in samples]
+            toks = tokenizer(captions)
+            batch["text"] = {"input_ids": torch.as_tensor(toks["input_ids"]),
+                             "attention_mask": torch.as_tensor(toks["attention_mask"])}
+        return batch
+
+    return collate
+
+
 class ShardedSampler(torch.utils.data.Sampler):
     """Shuffled, rank-sharded, drop-remainder sampler (the grain
     IndexSampler + ShardByJaxProcess equivalent, reference :299-305)."""
