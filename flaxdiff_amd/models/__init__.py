@@ -5,6 +5,7 @@ from .common import (Conv, ConvLayer, Dense, Downsample, FourierEmbedding,
 from .attention import (BasicTransformerBlock, EfficientAttention, FeedForward,
                         GEGLU, NormalAttention, TransformerBlock)
 from .unet import Unet
+from .unet_3d import UNet3D
 from .simple_dit import DiTBlock, SimpleDiT
 from .simple_mmdit import (HierarchicalMMDiT, MMAdaLNZero, MMDiTBlock,
                            PatchExpanding, PatchMerging, SimpleMMDiT)
@@ -16,7 +17,7 @@ from .vit_common import (AdaLNParams, AdaLNZero, PatchEmbedding,
                          apply_rotary_embedding)
 
 __all__ = [
-    "Unet", "Conv", "ConvLayer", "Dense", "Downsample", "FourierEmbedding",
+    "Unet", "UNet3D", "Conv", "ConvLayer", "Dense", "Downsample", "FourierEmbedding",
     "GroupNorm", "PixelShuffle", "RMSNorm", "ResidualBlock", "SeparableConv",
     "TimeEmbedding", "TimeProjection", "Upsample", "WeightStandardizedConv",
     "BasicTransformerBlock", "EfficientAttention", "FeedForward", "GEGLU",

@@ -269,6 +269,19 @@ class SimpleAutoEncoder(AutoEncoder):
         self.__downscale_factor__ = 2 ** (len(feature_depths) - 1)
         self.__latent_channels__ = latent_channels
 
+    def to(self, *args, **kwargs) -> "SimpleAutoEncoder":
+        """Move/cast the wrapped modules (not an nn.Module itself — the
+        AutoEncoder interface is functional, reference autoencoder.py:11)."""
+        for m in (self.encoder, self.decoder, self.quant_conv,
+                  self.post_quant_conv):
+            m.to(*args, **kwargs)
+        return self
+
+    def parameters(self):
+        for m in (self.encoder, self.decoder, self.quant_conv,
+                  self.post_quant_conv):
+            yield from m.parameters()
+
     def __encode__(self, x, key=None, **kw):
         h = self.quant_conv(self.encoder(x))
         mean, _ = h.chunk(2, dim=-1)

@@ -28,7 +28,7 @@ from .common import Dense, FourierEmbedding, TimeProjection
 from .hilbert import (build_2d_sincos_pos_embed, hilbert_indices,
                       hilbert_patchify, hilbert_unpatchify, zigzag_indices,
                       zigzag_patchify)
-from .vit_common import (AdaLNParams, PatchEmbedding, RoPEAttention,
+from .vit_common import (AdaLNParams, PatchEmbedding, RoPEAttention, norm_fp32,
                          RotaryEmbedding, layer_norm_noaffine, unpatchify)
 
 
@@ -160,7 +160,7 @@ class SimpleDiT(nn.Module):
         for block in self.blocks:
             tokens = block(tokens, cond, freqs_cis)
 
-        out = self.final_norm(tokens.float()).to(tokens.dtype)
+        out = norm_fp32(self.final_norm, tokens)
         out = self.final_proj(out)
 
         if self.learn_sigma:

@@ -28,7 +28,7 @@ from .common import Dense, FourierEmbedding, TimeProjection
 from .hilbert import (hilbert_indices, hilbert_patchify, hilbert_unpatchify,
                       inverse_permutation)
 from .simple_dit import DiTBlock  # noqa: F401  (re-export convenience)
-from .vit_common import (PatchEmbedding, RoPEAttention, RotaryEmbedding,
+from .vit_common import (PatchEmbedding, RoPEAttention, RotaryEmbedding, norm_fp32,
                          layer_norm_noaffine, unpatchify)
 
 
@@ -152,7 +152,7 @@ class SimpleMMDiT(nn.Module):
         for block in self.blocks:
             seq = block(seq, t_emb, text_emb, freqs_cis)
 
-        out = self.final_norm(seq.float()).to(seq.dtype)
+        out = norm_fp32(self.final_norm, seq)
         out = self.final_proj(out)
         if self.learn_sigma:
             out, _ = out.chunk(2, dim=-1)
@@ -179,7 +179,7 @@ class PatchMerging(nn.Module):
         x = x.reshape(B, h_p // m, m, w_p // m, m, C)
         x = x.permute(0, 1, 3, 2, 4, 5).reshape(B, (h_p // m) * (w_p // m),
                                                 m * m * C)
-        x = self.norm(x.float()).to(x.dtype)
+        x = norm_fp32(self.norm, x)
         x = self.projection(x)
         return x, h_p // m, w_p // m
 
@@ -200,7 +200,7 @@ class PatchExpanding(nn.Module):
         B, L, C = x.shape
         e = self.expand_size
         x = self.projection(x)
-        x = self.norm(x.float()).to(x.dtype)
+        x = norm_fp32(self.norm, x)
         x = x.reshape(B, h_p, w_p, e, e, self.out_features)
         x = x.permute(0, 1, 3, 2, 4, 5).reshape(B, (h_p * e) * (w_p * e),
                                                 self.out_features)
@@ -324,13 +324,13 @@ class HierarchicalMMDiT(nn.Module):
         for i, stage in enumerate(range(n - 2, -1, -1)):
             seq, ch, cw = self.patch_expanders[i](seq, ch, cw)
             seq = torch.cat([seq, skips[stage]], dim=-1)
-            seq = self.fusion_norms[i](seq.float()).to(seq.dtype)
+            seq = norm_fp32(self.fusion_norms[i], seq)
             seq = self.fusion_denses[i](seq)
             freqs = rope_for(stage, seq.shape[1], seq.device)
             for block in self.decoder_blocks[i]:
                 seq = block(seq, t_embs[stage], text_embs[stage], freqs)
 
-        out = self.final_norm(seq.float()).to(seq.dtype)
+        out = norm_fp32(self.final_norm, seq)
         out = self.final_proj(out)
         if self.learn_sigma:
             out, _ = out.chunk(2, dim=-1)

@@ -26,7 +26,7 @@ from .attention import TransformerBlock
 from .common import Conv, Dense, FourierEmbedding, TimeProjection
 from .hilbert import hilbert_patchify, hilbert_unpatchify
 from .simple_dit import DiTBlock
-from .vit_common import PatchEmbedding, RotaryEmbedding, unpatchify
+from .vit_common import PatchEmbedding, RotaryEmbedding, norm_fp32, unpatchify
 
 
 class UViT(nn.Module):
@@ -131,7 +131,7 @@ class UViT(nn.Module):
             seq = dense(seq)
             seq = blk(seq)
 
-        seq = self.final_norm(seq.float()).to(seq.dtype)
+        seq = norm_fp32(self.final_norm, seq)
         patches_out = self.final_proj(seq[:, :num_patches, :])
 
         if self.use_hilbert:
@@ -229,7 +229,7 @@ class SimpleUDiT(nn.Module):
             seq = dense(seq)
             seq = blk(seq, cond, None)
 
-        out = self.final_norm(seq.float()).to(seq.dtype)
+        out = norm_fp32(self.final_norm, seq)
         out = self.final_proj(out)
         if self.learn_sigma:
             out, _ = out.chunk(2, dim=-1)

@@ -32,7 +32,7 @@ from .hilbert import (build_2d_sincos_pos_embed, hilbert_indices,
                       hilbert_patchify, hilbert_unpatchify, inverse_permutation,
                       zigzag_indices, zigzag_patchify)
 from .simple_dit import DiTBlock
-from .vit_common import (AdaLNParams, PatchEmbedding, RotaryEmbedding,
+from .vit_common import (AdaLNParams, PatchEmbedding, RotaryEmbedding, norm_fp32,
                          layer_norm_noaffine, unpatchify)
 
 
@@ -327,7 +327,7 @@ class HybridSSMAttentionDiT(nn.Module):
         for block in self.blocks:
             seq = block(seq, cond, freqs)
 
-        out = self.final_norm(seq.float()).to(seq.dtype)
+        out = norm_fp32(self.final_norm, seq)
         out = self.final_proj(out)
         if self.learn_sigma:
             out, _ = out.chunk(2, dim=-1)

@@ -23,6 +23,16 @@ from .. import ops
 from .common import Dense
 
 
+def norm_fp32(norm: nn.LayerNorm, x: torch.Tensor) -> torch.Tensor:
+    """Run a LayerNorm in fp32 whatever the module/activation dtype (the
+    AdaLN pre-norms are fp32 for stability; after model.bfloat16() the params
+    are bf16, and torch rejects mixed-dtype layer_norm on CPU)."""
+    w = norm.weight.float() if norm.weight is not None else None
+    b = norm.bias.float() if norm.bias is not None else None
+    return nn.functional.layer_norm(
+        x.float(), norm.normalized_shape, w, b, norm.eps).to(x.dtype)
+
+
 def unpatchify(x: torch.Tensor, channels: int = 3) -> torch.Tensor:
     """[B, N, p*p*C] (square raster grid) -> NHWC image."""
     patch_size = int((x.shape[2] // channels) ** 0.5)

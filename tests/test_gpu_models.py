@@ -1,0 +1,93 @@
+"""GPU smoke tests: every model family runs forward+backward in bf16 on the
+HIP kernel set (the CPU suites cover numerics; these prove the native path
+executes for the full architecture zoo of SURVEY.md §2.3)."""
+import pytest
+import torch
+
+pytestmark = [pytest.mark.gpu,
+              pytest.mark.timeout(240, method="thread")]
+
+
+def _run(model, *inputs):
+    model = model.cuda().bfloat16()
+    outs = model(*[i.cuda().bfloat16() if torch.is_tensor(i) and i.is_floating_point()
+                   else i for i in inputs])
+    loss = outs.float().pow(2).mean()
+    loss.backward()
+    grads = [p.grad for p in model.parameters() if p.grad is not None]
+    assert grads, "no grads produced"
+    assert all(torch.isfinite(g.float()).all() for g in grads)
+    return outs
+
+
+def test_simple_dit_gpu():
+    from flaxdiff_amd.models import SimpleDiT
+    torch.manual_seed(0)
+    m = SimpleDiT(patch_size=4, emb_features=64, num_layers=2, num_heads=4,
+                  context_dim=32)
+    _run(m, torch.randn(2, 16, 16, 3), torch.rand(2), torch.randn(2, 7, 32))
+
+
+def test_simple_dit_hilbert_gpu():
+    from flaxdiff_amd.models import SimpleDiT
+    m = SimpleDiT(patch_size=4, emb_features=64, num_layers=2, num_heads=4,
+                  context_dim=32, use_hilbert=True)
+    _run(m, torch.randn(2, 16, 16, 3), torch.rand(2), torch.randn(2, 7, 32))
+
+
+def test_uvit_gpu():
+    from flaxdiff_amd.models import UViT
+    m = UViT(patch_size=4, emb_features=64, num_layers=4, num_heads=4,
+             context_dim=32)
+    _run(m, torch.randn(2, 16, 16, 3), torch.rand(2), torch.randn(2, 7, 32))
+
+
+def test_simple_udit_gpu():
+    from flaxdiff_amd.models import SimpleUDiT
+    m = SimpleUDiT(patch_size=4, emb_features=64, num_layers=4, num_heads=4,
+                   context_dim=32)
+    _run(m, torch.randn(2, 16, 16, 3), torch.rand(2), torch.randn(2, 7, 32))
+
+
+def test_simple_mmdit_gpu():
+    from flaxdiff_amd.models import SimpleMMDiT
+    m = SimpleMMDiT(patch_size=4, emb_features=64, num_layers=2, num_heads=4,
+                    context_dim=32)
+    _run(m, torch.randn(2, 16, 16, 3), torch.rand(2), torch.randn(2, 7, 32))
+
+
+def test_hierarchical_mmdit_gpu():
+    from flaxdiff_amd.models import HierarchicalMMDiT
+    m = HierarchicalMMDiT(base_patch_size=2, emb_features=(32, 48, 64),
+                          num_layers=(1, 1, 2), num_heads=(4, 4, 4),
+                          context_dim=32)
+    _run(m, torch.randn(2, 16, 16, 3), torch.rand(2), torch.randn(2, 7, 32))
+
+
+def test_hybrid_ssm_dit_gpu():
+    from flaxdiff_amd.models import HybridSSMAttentionDiT
+    m = HybridSSMAttentionDiT(patch_size=4, emb_features=64, num_layers=4,
+                              num_heads=4, ssm_state_dim=8, context_dim=32)
+    _run(m, torch.randn(2, 16, 16, 3), torch.rand(2), torch.randn(2, 7, 32))
+
+
+def test_unet3d_gpu():
+    from flaxdiff_amd.models import UNet3D
+    m = UNet3D(emb_features=32, feature_depths=(16, 32),
+               attention_configs=({"heads": 2}, {"heads": 2}),
+               num_res_blocks=1, norm_groups=4, context_dim=16)
+    _run(m, torch.randn(2, 3, 16, 16, 3), torch.rand(2), torch.randn(2, 5, 16))
+
+
+def test_autoencoder_gpu():
+    from flaxdiff_amd.models.autoencoder import SimpleAutoEncoder
+    torch.manual_seed(0)
+    ae = SimpleAutoEncoder(latent_channels=4,
+                           feature_depths=(16, 32)).to("cuda", torch.bfloat16)
+    x = torch.randn(2, 16, 16, 3, device="cuda").bfloat16()
+    lat = ae.encode(x)
+    rec = ae.decode(lat)
+    assert rec.shape == x.shape
+    rec.float().pow(2).mean().backward()
+    assert all(torch.isfinite(p.grad.float()).all() for p in ae.parameters()
+               if p.grad is not None)
