@@ -19,7 +19,12 @@ from ..utils import RandomMarkovState
 from .utils import load_from_checkpoint, parse_config
 
 
-class DiffusionInferencePipeline:
+class InferencePipeline:
+    """Generic base marker (reference pipeline.py:26-40 — a thin base the
+    diffusion pipeline subclasses)."""
+
+
+class DiffusionInferencePipeline(InferencePipeline):
     def __init__(self, model, noise_schedule, prediction_transform,
                  input_config=None, autoencoder=None, checkpoint: Optional[dict] = None,
                  config: Optional[dict] = None, step: Optional[int] = None):
