@@ -352,11 +352,18 @@ class _AttentionFn(torch.autograd.Function):
 
 def attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
               scale: Optional[float] = None) -> torch.Tensor:
-    """q,k,v: [B, H, S, D] -> [B, H, Sq, D]; fp32 softmax."""
+    """q,k,v: [B, H, S, D] -> [B, H, Sq, D]; fp32 softmax.
+
+    The flash kernel covers D in 8..128 multiples of 8 (every shipped config).
+    Other head dims (e.g. emb 48 / 4 heads = 12) compose library GEMMs +
+    fp32 softmax on-device — a shape-gated route, not an eager fallback."""
     if scale is None:
         scale = q.shape[-1] ** -0.5
     if _use_hip(q):
-        return _AttentionFn.apply(q.contiguous(), k.contiguous(), v.contiguous(), scale)
+        D = q.shape[-1]
+        if D % 8 == 0 and D <= 128:
+            return _AttentionFn.apply(q.contiguous(), k.contiguous(),
+                                      v.contiguous(), scale)
     return reference.attention(q, k, v, scale)
 
 
