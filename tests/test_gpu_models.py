@@ -91,3 +91,25 @@ def test_autoencoder_gpu():
     rec.float().pow(2).mean().backward()
     assert all(torch.isfinite(p.grad.float()).all() for p in ae.parameters()
                if p.grad is not None)
+
+
+def test_dit_train_step_gpu(tmp_path):
+    """SimpleDiT through the full trainer (fused optimizer, bf16 shadow) on
+    the HIP path."""
+    import math
+    from flaxdiff_amd.predictors import KarrasPredictionTransform
+    from flaxdiff_amd.schedulers import EDMNoiseScheduler
+    from flaxdiff_amd.trainer import DiffusionTrainer
+    from flaxdiff_amd.models import SimpleDiT
+
+    model = SimpleDiT(patch_size=4, emb_features=64, num_layers=2,
+                      num_heads=4, context_dim=768)
+    trainer = DiffusionTrainer(
+        model, EDMNoiseScheduler(1, sigma_max=80),
+        KarrasPredictionTransform(sigma_data=0.5),
+        name="dit-gpu", checkpoint_base_path=str(tmp_path),
+        compute_dtype=torch.bfloat16, distributed=False)
+    batch = {"image": torch.randint(0, 255, (4, 16, 16, 3), dtype=torch.uint8)}
+    out1 = trainer.train_step(batch)
+    out2 = trainer.train_step(batch)
+    assert math.isfinite(out1["loss"]) and math.isfinite(out2["loss"])
