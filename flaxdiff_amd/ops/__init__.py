@@ -258,20 +258,54 @@ def conv2d(x: torch.Tensor, w: torch.Tensor, b: Optional[torch.Tensor] = None,
     return reference.conv2d_nhwc(x, w, b, stride=stride, padding="same")
 
 
+class _ConvTransposeFn(torch.autograd.Function):
+    """stride-2 transposed conv as ONE dgrad-kernel call.
+
+    convT(x, w) is exactly the input-gradient of the stride-2 SAME conv with
+    weight wf = w.permute(0,1,3,2) (verified bit-exact against the
+    zero-stuff + flipped-conv form on CPU, incl. odd sizes) — so the forward
+    runs the implicit-GEMM dgrad kernel directly with no zero-stuffed pixels
+    (the former zero-stuff path spent 3/4 of its MACs on zeros). Adjoints:
+    d/dx = the forward conv, d/dwf = the split-M wgrad with roles swapped.
+    """
+
+    @staticmethod
+    def forward(ctx, x, wf, stride):
+        ext = _require_ext()
+        H2 = x.shape[1] * stride
+        W2 = x.shape[2] * stride
+        y = ext.conv2d_dgrad(x, wf, stride, H2, W2)
+        ctx.save_for_backward(x, wf)
+        ctx.stride = stride
+        return y
+
+    @staticmethod
+    def backward(ctx, dout):
+        x, wf = ctx.saved_tensors
+        ext = _require_ext()
+        dout = dout.contiguous()
+        dx = ext.conv2d_fwd(dout, wf, torch.Tensor(), ctx.stride) \
+            if ctx.needs_input_grad[0] else None
+        dwf = None
+        if ctx.needs_input_grad[1]:
+            dwf, _ = ext.conv2d_wgrad(x, dout, wf.shape[0], wf.shape[1],
+                                      ctx.stride)
+            dwf = dwf.to(wf.dtype)
+        return dx, dwf, None
+
+
 def conv2d_transpose(x: torch.Tensor, w: torch.Tensor, b: Optional[torch.Tensor] = None,
                      stride: int = 2) -> torch.Tensor:
-    """flax-SAME transposed conv. On GPU: zero-stuff + spatially-flipped
-    SAME conv through the native implicit-GEMM kernel (convT(x) ==
-    conv(pad(zerostuff(x)), flip(w)) cropped to H*stride)."""
+    """flax-SAME transposed conv; w is [kh, kw, Cin, Cout]. On GPU the
+    stride-2 3x3 case runs the implicit-GEMM dgrad kernel directly
+    (_ConvTransposeFn)."""
     kh, kw = w.shape[0], w.shape[1]
     if _use_hip(x) and stride == 2 and kh == 3 and kw == 3:
-        B, H, W, Ci = x.shape
-        stuffed = torch.zeros(B, 2 * H + 1, 2 * W + 1, Ci, device=x.device,
-                              dtype=x.dtype)
-        stuffed[:, 1::2, 1::2] = x
-        wf = torch.flip(w, dims=(0, 1)).contiguous()
-        out = conv2d(stuffed, wf, b, stride=1)
-        return out[:, : 2 * H, : 2 * W].contiguous()
+        wf = w.permute(0, 1, 3, 2).contiguous()
+        out = _ConvTransposeFn.apply(x.contiguous(), wf, stride)
+        if b is not None:
+            out = out + b
+        return out
     return reference.conv2d_transpose_nhwc(x, w, b, stride)
 
 
