@@ -119,3 +119,20 @@ def test_dense_flax_convention():
     y = d(x)
     assert y.shape == (3, 16)
     assert torch.allclose(y, x @ d.weight + d.bias)
+
+
+def test_conv_transpose_equals_conv_input_grad():
+    """The GPU conv_transpose path runs the dgrad kernel directly; lock the
+    underlying identity on CPU: convT(x, w) == d/d(input) of the stride-2
+    SAME conv with channel-transposed weight (ops/__init__._ConvTransposeFn)."""
+    from flaxdiff_amd.ops import reference
+    torch.manual_seed(0)
+    for (H, W, Ci, Co) in [(8, 8, 16, 24), (7, 5, 8, 8)]:
+        x = torch.randn(2, H, W, Ci)
+        w = torch.randn(3, 3, Ci, Co) * 0.1
+        ref = reference.conv2d_transpose_nhwc(x, w, None, 2)
+        big = torch.zeros(2, 2 * H, 2 * W, Co, requires_grad=True)
+        wf = w.permute(0, 1, 3, 2).contiguous()
+        y = reference.conv2d_nhwc(big, wf, None, stride=2)
+        g = torch.autograd.grad(y, big, grad_outputs=x)[0]
+        assert torch.equal(ref, g)
