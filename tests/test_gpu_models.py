@@ -134,7 +134,9 @@ def test_all_samplers_gpu():
                norm_groups=8).cuda().bfloat16().eval()
 
     def model(x, t, *cond):
-        return net(x.bfloat16(), t).float()
+        ctx = torch.zeros(x.shape[0], 77, 768, device=x.device,
+                          dtype=torch.bfloat16)
+        return net(x.bfloat16(), t, ctx).float()
 
     from flaxdiff_amd.predictors import EpsilonPredictionTransform
     from flaxdiff_amd.schedulers import CosineNoiseScheduler
