@@ -204,3 +204,29 @@ def test_autoencoder_trainer_reconstruction(tmp_path):
     losses = [tr.train_step(batch)["rec_loss"] for _ in range(30)]
     assert losses[-1] < losses[0] * 0.9
     assert all(l == l for l in losses)
+
+
+def test_resume_continues_identically(tmp_path):
+    """Interrupted training must continue exactly where it left off: 3 steps
+    + save + reload + 2 steps == 5 uninterrupted steps (weights, EMA and
+    optimizer moments all bit-equal — RNG chain included in the checkpoint)."""
+    batch = {"image": torch.randint(
+        0, 255, (4, 16, 16, 3), dtype=torch.uint8,
+        generator=torch.Generator().manual_seed(11))}
+    it = iter(lambda: batch, None)
+
+    torch.manual_seed(0)
+    straight = tiny_trainer(tmp_path / "a")
+    straight.train_loop(it, steps=5)
+
+    torch.manual_seed(0)
+    part1 = tiny_trainer(tmp_path / "b")
+    part1.train_loop(it, steps=3)
+    part1.save(block=True)
+    resumed = tiny_trainer(tmp_path / "b", load_from_checkpoint=True)
+    resumed.train_loop(it, steps=2)
+
+    assert resumed.global_step == straight.global_step == 5
+    assert torch.equal(resumed.optimizer.flat, straight.optimizer.flat)
+    assert torch.equal(resumed.optimizer.ema, straight.optimizer.ema)
+    assert torch.equal(resumed.optimizer.exp_avg, straight.optimizer.exp_avg)
