@@ -151,3 +151,69 @@ def test_data_parallel_text_conditional_cfg():
         res = dict(results)
     assert res[0]["params_equal"] and res[1]["params_equal"]
     assert res[0]["losses"] == pytest.approx(res[1]["losses"], rel=1e-5)
+
+
+def _resume_worker(rank, world, port, ckpt_dir, results):
+    os.environ.update({
+        "RANK": str(rank), "WORLD_SIZE": str(world), "LOCAL_RANK": str(rank),
+        "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+    })
+    import torch.distributed as dist
+    from flaxdiff_amd.predictors import EpsilonPredictionTransform
+    from flaxdiff_amd.schedulers import CosineNoiseScheduler
+    from flaxdiff_amd.trainer import DiffusionTrainer
+
+    def make():
+        torch.manual_seed(1234)
+        m = Unet(emb_features=32, feature_depths=[8, 16],
+                 attention_configs=[None, {"heads": 2}], num_res_blocks=1,
+                 num_middle_res_blocks=1, norm_groups=4, context_dim=16)
+        return m
+
+    g = torch.Generator().manual_seed(100 + rank)
+    batch = {"image": torch.randint(0, 255, (4, 16, 16, 3), generator=g,
+                                    dtype=torch.uint8)}
+    it = iter(lambda: batch, None)
+
+    tr = DiffusionTrainer(make(), CosineNoiseScheduler(1000),
+                          EpsilonPredictionTransform(), name="ddp-resume",
+                          checkpoint_base_path=ckpt_dir,
+                          text_context_shape=(4, 16), distributed=True)
+    tr.train_loop(it, steps=2)
+    tr.save(block=True)
+    dist.barrier()
+
+    tr2 = DiffusionTrainer(make(), CosineNoiseScheduler(1000),
+                           EpsilonPredictionTransform(), name="ddp-resume",
+                           checkpoint_base_path=ckpt_dir,
+                           text_context_shape=(4, 16), distributed=True,
+                           load_from_checkpoint=True)
+    tr2.train_loop(it, steps=1)
+    # ranks agree after resume + one more synced step
+    flat = tr2.optimizer.flat.clone()
+    gathered = [torch.zeros_like(flat) for _ in range(world)]
+    dist.all_gather(gathered, flat)
+    same = all(torch.equal(gathered[0], gi) for gi in gathered)
+    results[rank] = {"step": tr2.global_step, "params_equal": bool(same)}
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_data_parallel_resume(tmp_path):
+    """Rank-0 checkpoints + resume keep all ranks in lockstep (the driver's
+    multi-GPU runs restart from rank-0 checkpoints)."""
+    world = 2
+    ctx = mp.get_context("spawn")
+    with ctx.Manager() as man:
+        results = man.dict()
+        procs = [ctx.Process(target=_resume_worker,
+                             args=(r, world, 29741, str(tmp_path), results))
+                 for r in range(world)]
+        for p in procs:
+            p.start()
+        for p in procs:
+            p.join(240)
+            assert p.exitcode == 0
+        for r in range(world):
+            assert results[r]["step"] == 3
+            assert results[r]["params_equal"]
