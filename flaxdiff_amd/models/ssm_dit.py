@@ -27,6 +27,7 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
+from .. import ops
 from .common import Dense, FourierEmbedding, TimeProjection
 from .hilbert import (build_2d_sincos_pos_embed, hilbert_indices,
                       hilbert_patchify, hilbert_unpatchify, inverse_permutation,
@@ -134,15 +135,14 @@ class SpatialFusionConv(nn.Module):
             for _ in dilations])
 
     def forward(self, y_2d: torch.Tensor) -> torch.Tensor:
-        # NHWC -> NCHW for torch depthwise; zero-init so this path is cold at
-        # start. (A fused NHWC dilated-depthwise HIP kernel is a follow-up.)
-        x = y_2d.permute(0, 3, 1, 2)
-        out = x
-        C = x.shape[1]
+        # NHWC throughout: the dilated depthwise runs on the HIP kernels of
+        # depthwise.hip on GPU (weights stay in the torch (C,1,k,k) layout
+        # for state-dict stability; permuted to [k,k,C,1] at call).
+        out = y_2d
         for w, dil in zip(self.weights, self.dilations):
-            pad = dil * (self.kernel_size - 1) // 2
-            out = out + F.conv2d(x, w.to(x.dtype), None, 1, pad, dil, groups=C)
-        return out.permute(0, 2, 3, 1)
+            wk = w.permute(2, 3, 0, 1).to(y_2d.dtype)
+            out = out + ops.depthwise_conv2d(y_2d, wk, None, 1, dil)
+        return out
 
 
 class SSMDiTBlock(nn.Module):
