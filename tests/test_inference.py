@@ -121,3 +121,35 @@ def test_pipeline_from_checkpoint(tmp_path):
     s2 = pipe.get_sampler(EulerAncestralSampler, 0.0)
     assert s1 is s2
     assert pipe.get_sampler(EulerAncestralSampler, 2.0) is not s1
+
+
+@pytest.mark.parametrize("arch,model_cfg", [
+    ("unet", {"emb_features": 32, "feature_depths": [8, 16],
+              "attention_configs": [None, {"heads": 2}], "num_res_blocks": 1,
+              "norm_groups": 4, "context_dim": 16}),
+    ("uvit", {"patch_size": 4, "emb_features": 32, "num_layers": 2,
+              "num_heads": 2, "context_dim": 16}),
+    ("simple_dit", {"patch_size": 4, "emb_features": 32, "num_layers": 1,
+                    "num_heads": 2, "context_dim": 16}),
+    ("simple_udit", {"patch_size": 4, "emb_features": 32, "num_layers": 2,
+                     "num_heads": 2, "context_dim": 16}),
+    ("simple_mmdit", {"patch_size": 4, "emb_features": 32, "num_layers": 1,
+                      "num_heads": 2, "context_dim": 16}),
+    ("hierarchical_mmdit", {"base_patch_size": 2,
+                            "emb_features": (16, 24, 32),
+                            "num_layers": (1, 1, 1), "num_heads": (2, 2, 2),
+                            "context_dim": 16}),
+    ("hybrid_dit", {"patch_size": 4, "emb_features": 32, "num_layers": 2,
+                    "num_heads": 2, "ssm_state_dim": 4, "context_dim": 16}),
+])
+def test_parse_config_every_architecture(arch, model_cfg):
+    """Every registry architecture reconstructs from a run manifest and runs
+    one forward pass (reference utils.py:120-134 registry)."""
+    out = parse_config({"architecture": arch, "model": model_cfg,
+                        "noise_schedule": "edm",
+                        "arguments": {"image_size": 16}})
+    model = out["model"]
+    y = model(torch.randn(1, 16, 16, 3), torch.rand(1),
+              torch.randn(1, 7, 16))
+    assert y.shape[0] == 1
+    assert torch.isfinite(y).all()
