@@ -1,0 +1,52 @@
+#!/usr/bin/env python3
+"""Latent diffusion: train the UNet in an autoencoder's latent space
+(the reference's LDM config — BASELINE config 5 shape)."""
+import argparse
+import os
+import sys
+
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
+
+import torch
+
+from flaxdiff_amd.models import Unet
+from flaxdiff_amd.models.autoencoder import SimpleAutoEncoder
+from flaxdiff_amd.predictors import KarrasPredictionTransform
+from flaxdiff_amd.schedulers import EDMNoiseScheduler
+from flaxdiff_amd.trainer import DiffusionTrainer
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--steps", type=int, default=5)
+    ap.add_argument("--size", type=int, default=32)
+    args = ap.parse_args()
+
+    dev = "cuda" if torch.cuda.is_available() else "cpu"
+    dt = torch.bfloat16 if dev == "cuda" else torch.float32
+    ae = SimpleAutoEncoder(latent_channels=4,
+                           feature_depths=(16, 32)).to(dev, dt)
+    model = Unet(in_channels=4, output_channels=4, emb_features=32,
+                 feature_depths=[8, 16],
+                 attention_configs=[None, {"heads": 2}], num_res_blocks=1,
+                 norm_groups=4, context_dim=16)
+    trainer = DiffusionTrainer(
+        model, EDMNoiseScheduler(1, sigma_max=80, sigma_data=0.5),
+        KarrasPredictionTransform(sigma_data=0.5),
+        name="ldm", checkpoint_base_path="./checkpoints",
+        autoencoder=ae, text_context_shape=(4, 16),
+        compute_dtype=dt, distributed=False)
+
+    def batches():
+        g = torch.Generator().manual_seed(0)
+        while True:
+            yield {"image": torch.randint(
+                0, 255, (4, args.size, args.size, 3), generator=g,
+                dtype=torch.uint8)}
+
+    trainer.train_loop(batches(), steps=args.steps)
+    print("ldm trained", args.steps, "steps; final loss ok")
+
+
+if __name__ == "__main__":
+    main()

@@ -137,6 +137,11 @@ class DiffusionTrainer(SimpleTrainer):
                 autoencoder=self.autoencoder)
             cond = (conditioning_context,) if conditioning_context is not None else \
                 (self.null_context.unsqueeze(0).expand(num_samples, *self.null_context.shape),)
+            if conditioning_context is not None and sampler.guidance_scale > 0 \
+                    and not sampler.unconditionals:
+                # CFG needs a null embedding per condition; the trainer's
+                # null context is the uncond row (reference :141-148)
+                sampler.unconditionals = [self.null_context]
             samples = sampler.generate_samples(
                 num_samples=num_samples, resolution=resolution,
                 diffusion_steps=diffusion_steps,

@@ -230,3 +230,18 @@ def test_resume_continues_identically(tmp_path):
     assert torch.equal(resumed.optimizer.flat, straight.optimizer.flat)
     assert torch.equal(resumed.optimizer.ema, straight.optimizer.ema)
     assert torch.equal(resumed.optimizer.exp_avg, straight.optimizer.exp_avg)
+
+
+def test_validation_sample_with_cfg(tmp_path):
+    """validation_sample with conditioning + guidance>0 must supply the null
+    embedding for the CFG uncond half (fixed: sampler got no unconditionals
+    when built without an input_config)."""
+    torch.manual_seed(0)
+    tr = tiny_trainer(tmp_path)
+    from flaxdiff_amd.samplers import EulerAncestralSampler
+    ctx = torch.randn(2, 4, 16)
+    out = tr.validation_sample(EulerAncestralSampler, num_samples=2,
+                               resolution=16, diffusion_steps=2,
+                               guidance_scale=3.0, conditioning_context=ctx)
+    assert out.shape == (2, 16, 16, 3)
+    assert torch.isfinite(out).all()
