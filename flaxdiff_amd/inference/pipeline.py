@@ -16,7 +16,7 @@ import torch
 
 from ..samplers import DiffusionSampler, EulerAncestralSampler
 from ..utils import RandomMarkovState
-from .utils import load_from_checkpoint, parse_config
+from .utils import load_from_checkpoint
 
 
 class InferencePipeline:

@@ -7,7 +7,7 @@ fixed-dimension embedding for tests and synthetic benchmarks.
 """
 from __future__ import annotations
 
-from typing import Any, List, Optional
+from typing import List, Optional
 
 import torch
 

@@ -11,7 +11,6 @@ tower), matching common FID-variant practice.
 """
 from __future__ import annotations
 
-import math
 from typing import Callable, Optional
 
 import numpy as np

@@ -12,8 +12,6 @@ checkpoint path as `modelname`.
 """
 from __future__ import annotations
 
-import math
-from typing import Optional
 
 import torch
 import torch.nn.functional as F

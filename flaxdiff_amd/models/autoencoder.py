@@ -18,15 +18,13 @@ local diffusers-format safetensors/pt file via `load_diffusers_weights`
 from __future__ import annotations
 
 import abc
-import math
 from typing import Optional, Sequence
 
 import torch
 import torch.nn as nn
-import torch.nn.functional as F
 
 from .attention import NormalAttention
-from .common import Conv, GroupNorm, _cast
+from .common import Conv, GroupNorm
 from .. import ops
 
 

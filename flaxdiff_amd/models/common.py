@@ -13,7 +13,7 @@ dtype, so bf16 training accumulates gradients into fp32 masters.
 from __future__ import annotations
 
 import math
-from typing import Callable, Optional, Sequence
+from typing import Callable, Optional
 
 import torch
 import torch.nn as nn

@@ -17,7 +17,7 @@ causal path is a prefix-sum over the sequence.
 from __future__ import annotations
 
 import math
-from typing import Callable, Optional
+from typing import Optional
 
 import torch
 import torch.nn as nn

@@ -21,7 +21,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from .attention import TransformerBlock
-from .common import (Conv, ConvLayer, Downsample, FourierEmbedding, GroupNorm,
+from .common import (ConvLayer, Downsample, FourierEmbedding, GroupNorm,
                      RMSNorm, ResidualBlock, TimeProjection, Upsample)
 
 

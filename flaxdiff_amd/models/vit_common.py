@@ -13,7 +13,6 @@ ops.attention.
 """
 from __future__ import annotations
 
-import math
 from typing import Optional, Tuple
 
 import torch

@@ -1,5 +1,4 @@
 """Euler-family ODE/SDE samplers (reference: samplers/euler.py:6-55)."""
-import torch
 
 from ..utils import RandomMarkovState
 from .common import DiffusionSampler

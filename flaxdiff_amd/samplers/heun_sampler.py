@@ -1,5 +1,4 @@
 """Heun 2nd-order sampler — 2 NFE/step (reference: samplers/heun_sampler.py:6-27)."""
-import torch
 
 from ..utils import RandomMarkovState
 from .common import DiffusionSampler

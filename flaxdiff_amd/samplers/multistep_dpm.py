@@ -1,5 +1,4 @@
 """Multistep DPM sampler with host-side history (reference: samplers/multistep_dpm.py:6-58)."""
-import torch
 
 from ..utils import RandomMarkovState
 from .common import DiffusionSampler

@@ -6,7 +6,7 @@ tables. Rate lookups return fp32 tensors broadcastable against NHWC batches.
 """
 from __future__ import annotations
 
-from typing import Tuple, Union
+from typing import Tuple
 
 import torch
 

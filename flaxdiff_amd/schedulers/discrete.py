@@ -7,7 +7,6 @@ moved to device on first use — the GPU train step then does a pure gather
 """
 from __future__ import annotations
 
-from typing import Tuple
 
 import numpy as np
 import torch

@@ -12,8 +12,7 @@ Adam+EMA one fused HIP pass. l2 here is 0.5*(x-y)^2 (optax.l2_loss).
 """
 from __future__ import annotations
 
-import math
-from typing import Callable, Dict, Optional
+from typing import Dict, Optional
 
 import torch
 

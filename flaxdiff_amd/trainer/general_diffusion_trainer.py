@@ -9,13 +9,10 @@ becomes a local "export best checkpoint" operation (no hard wandb dep).
 """
 from __future__ import annotations
 
-import math
-import time
 from typing import Dict, List, Optional
 
 import torch
 
-from ..utils import get_coeff_shapes_tuple
 from .diffusion_trainer import DiffusionTrainer
 
 

@@ -13,7 +13,7 @@ from __future__ import annotations
 
 import math
 import time
-from typing import Any, Callable, Dict, Optional, Tuple
+from typing import Callable, Dict, Optional
 
 import torch
 

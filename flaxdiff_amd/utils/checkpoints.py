@@ -12,7 +12,6 @@ AsyncCheckpointer equivalent).
 from __future__ import annotations
 
 import json
-import os
 import shutil
 import threading
 from pathlib import Path
