@@ -179,7 +179,8 @@ class Conv(nn.Module):
     def __init__(self, in_features: int, features: int, kernel_size=(3, 3),
                  strides=(1, 1), use_bias: bool = True):
         super().__init__()
-        kh, kw = kernel_size if isinstance(kernel_size, (tuple, list)) else (kernel_size, kernel_size)
+        kh, kw = kernel_size if isinstance(kernel_size, (tuple, list)) \
+            else (kernel_size, kernel_size)
         self.stride = strides[0] if isinstance(strides, (tuple, list)) else strides
         w = torch.empty(kh, kw, in_features, features)
         variance_scaling_(w, kh * kw * in_features, kh * kw * features)

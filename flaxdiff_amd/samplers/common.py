@@ -187,7 +187,8 @@ class DiffusionSampler:
         if rngstate is None:
             rngstate = RandomMarkovState(42)
         if device is None:
-            device = next(self.model.parameters()).device if hasattr(self.model, "parameters") else "cpu"
+            device = (next(self.model.parameters()).device
+                      if hasattr(self.model, "parameters") else "cpu")
         if start_step is None:
             start_step = self.noise_schedule.max_timesteps
             if isinstance(start_step, float):

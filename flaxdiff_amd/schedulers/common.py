@@ -64,7 +64,8 @@ class NoiseScheduler:
         return signal_rates.to(images.dtype) * images + noise_rates.to(images.dtype) * noise
 
     def remove_all_noise(self, noisy_images, noise, steps, clip_denoised=True, rates=None):
-        signal_rates, noise_rates = self.get_rates(steps, shape=get_coeff_shapes_tuple(noisy_images))
+        signal_rates, noise_rates = self.get_rates(
+            steps, shape=get_coeff_shapes_tuple(noisy_images))
         return (noisy_images - noise * noise_rates) / signal_rates
 
     def transform_inputs(self, x, steps):

@@ -39,7 +39,8 @@ class CosineGeneralNoiseScheduler(GeneralizedNoiseScheduler):
     def get_sigmas(self, steps):
         if not torch.is_tensor(steps):
             steps = torch.as_tensor(steps, dtype=torch.float32)
-        return torch.tan(self.theta_min + steps.float() * (self.theta_max - self.theta_min)) / self.kappa
+        return torch.tan(self.theta_min
+                         + steps.float() * (self.theta_max - self.theta_min)) / self.kappa
 
 
 class CosineContinuousNoiseScheduler(ContinuousNoiseScheduler):

@@ -45,10 +45,14 @@ class DiscreteNoiseScheduler(NoiseScheduler):
             "posterior_variance": posterior_variance,
             "posterior_log_variance_clipped": np.log(np.maximum(posterior_variance, 1e-20)),
             "posterior_mean_coef1": betas * np.sqrt(alpha_cumprod_prev) / (1 - alpha_cumprod),
-            "posterior_mean_coef2": (1 - alpha_cumprod_prev) * np.sqrt(alphas) / (1 - alpha_cumprod),
-            "p2_loss_weights": (p2_loss_weight_k + alpha_cumprod / (1 - alpha_cumprod)) ** (-p2_loss_weight_gamma),
+            "posterior_mean_coef2":
+                (1 - alpha_cumprod_prev) * np.sqrt(alphas) / (1 - alpha_cumprod),
+            "p2_loss_weights":
+                (p2_loss_weight_k + alpha_cumprod / (1 - alpha_cumprod))
+                ** (-p2_loss_weight_gamma),
         }
-        self._tables = {k: torch.from_numpy(np.ascontiguousarray(v)).float() for k, v in tables.items()}
+        self._tables = {k: torch.from_numpy(np.ascontiguousarray(v)).float()
+                        for k, v in tables.items()}
         self._tables_device = torch.device("cpu")
 
     # -- device management ---------------------------------------------------

@@ -5,5 +5,6 @@ from .general_diffusion_trainer import GeneralDiffusionTrainer, generate_modelna
 from .autoencoder_trainer import AutoEncoderTrainer
 
 __all__ = ["FlatAdamWEMA", "warmup_cosine_schedule", "SimpleTrainer",
-           "DiffusionTrainer", "GeneralDiffusionTrainer", "generate_modelname", "AutoEncoderTrainer",
+           "DiffusionTrainer", "GeneralDiffusionTrainer", "generate_modelname",
+           "AutoEncoderTrainer",
            "l2_loss"]

@@ -83,7 +83,8 @@ class RandomMarkovState(MarkovState):
         return RandomMarkovState(_splitmix64(self.seed ^ _splitmix64(data + 0xDEADBEEF)))
 
 
-def clip_images(images: torch.Tensor, clip_min: float = -1.0, clip_max: float = 1.0) -> torch.Tensor:
+def clip_images(images: torch.Tensor, clip_min: float = -1.0,
+                clip_max: float = 1.0) -> torch.Tensor:
     """Reference: flaxdiff/utils.py clip_images."""
     return torch.clamp(images, clip_min, clip_max)
 
