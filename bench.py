@@ -26,6 +26,8 @@ def main():
     ap.add_argument("--global-batch", type=int, default=256)
     ap.add_argument("--resolution", type=int, default=64)
     ap.add_argument("--dtype", type=str, default="bf16", choices=["bf16", "fp32"])
+    ap.add_argument("--profile", action="store_true",
+                    help="per-step hipEvent timings to stderr (SURVEY §5.1)")
     args = ap.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -83,12 +85,31 @@ def main():
         torch.cuda.synchronize()
     t0 = time.perf_counter()
     loss = 0.0
-    for _ in range(args.steps):
-        loss = trainer.train_step(batch)["loss"]
+    ev = []
+    if args.profile and use_gpu:
+        for _ in range(args.steps):
+            a = torch.cuda.Event(enable_timing=True)
+            b = torch.cuda.Event(enable_timing=True)
+            a.record()
+            loss = trainer.train_step(batch)["loss"]
+            b.record()
+            ev.append((a, b))
+    else:
+        for _ in range(args.steps):
+            loss = trainer.train_step(batch)["loss"]
     parallel.barrier()
     if use_gpu:
         torch.cuda.synchronize()
     elapsed = time.perf_counter() - t0
+    if ev:
+        import sys
+        times = [a.elapsed_time(b) for a, b in ev]
+        mean = sum(times) / len(times)
+        var = sum((x - mean) ** 2 for x in times) / len(times)
+        print(json.dumps({"profile": {"per_step_ms": [round(x, 3) for x in times],
+                                      "mean_ms": round(mean, 3),
+                                      "std_ms": round(var ** 0.5, 3),
+                                      "rank": rank}}), file=sys.stderr)
 
     # max over ranks
     el = torch.tensor([elapsed], dtype=torch.float64)
