@@ -49,7 +49,7 @@ def main():
         else torch.float32,
         distributed=False)
 
-    loader = get_dataset(f"synthetic-{64 if not args.tiny else 64}",
+    loader = get_dataset("synthetic-64",
                          global_batch_size=batch, image_size=size,
                          worker_count=0)
     it = iter(loader)
