@@ -35,7 +35,8 @@ def main():
         KarrasPredictionTransform(sigma_data=0.5),
         name="ldm", checkpoint_base_path="./checkpoints",
         autoencoder=ae, text_context_shape=(4, 16),
-        compute_dtype=dt, distributed=False)
+        compute_dtype=dt,
+        distributed=int(os.environ.get("WORLD_SIZE", "1")) > 1)
 
     def batches():
         g = torch.Generator().manual_seed(0)

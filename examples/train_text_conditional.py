@@ -54,7 +54,7 @@ def main():
         checkpoint_base_path="./checkpoints",
         compute_dtype=torch.bfloat16 if torch.cuda.is_available()
         else torch.float32,
-        distributed=False)
+        distributed=int(os.environ.get("WORLD_SIZE", "1")) > 1)
 
     captions = ["a red square", "a blue circle", "green noise", "white"]
     toks = encoder.tokenize(captions[: args.batch])

@@ -47,7 +47,7 @@ def main():
         name="uncond64", checkpoint_base_path="./checkpoints",
         compute_dtype=torch.bfloat16 if torch.cuda.is_available()
         else torch.float32,
-        distributed=False)
+        distributed=int(os.environ.get("WORLD_SIZE", "1")) > 1)
 
     loader = get_dataset("synthetic-64",
                          global_batch_size=batch, image_size=size,
