@@ -488,14 +488,18 @@ def s5_scan(a: torch.Tensor, bu: torch.Tensor) -> torch.Tensor:
 
 def fused_adamw_ema(params_f32, grads, exp_avg, exp_avg_sq, ema_f32, params_bf16,
                     *, lr, beta1, beta2, eps, weight_decay, step, ema_decay,
-                    grad_scale: float = 1.0):
+                    grad_scale: float = 1.0, scale_dev=None, skip_ctr=None):
     """In-place AdamW + EMA lerp over flat fp32 master buffers.
 
     grads may be bf16 (gets scaled by grad_scale, e.g. 1/world_size folded in).
     params_bf16 is refreshed from the fp32 master in the same pass (may be None).
+    scale_dev (optional fp32 device scalar) overrides grad_scale — built
+    on-device as grad_scale * clip-factor * finite-gate so the optimizer step
+    needs no GPU→host sync; a nonfinite/zero value skips the whole update and
+    bumps skip_ctr (optional int32 device scalar).
     """
     ext = _require_ext()
     ext.fused_adamw_ema(params_f32, grads, exp_avg, exp_avg_sq, ema_f32,
                         params_bf16 if params_bf16 is not None else torch.Tensor(),
                         lr, beta1, beta2, eps, weight_decay, step, ema_decay,
-                        grad_scale)
+                        grad_scale, scale_dev, skip_ctr)

@@ -47,6 +47,18 @@ class DiffusionSampler:
 
         self._graph_capture = False
         self._graphed: Optional["_GraphedSampleModel"] = None
+
+    # _uncond_cache holds device-resident copies of the null embeddings (an
+    # H2D copy inside hipGraph capture would be illegal). Reassigning
+    # unconditionals (DiffusionTrainer.validation_sample does this after
+    # construction) must invalidate the cache, so it is a managed property.
+    @property
+    def unconditionals(self):
+        return self._unconditionals
+
+    @unconditionals.setter
+    def unconditionals(self, value):
+        self._unconditionals = value
         self._uncond_cache = {}
 
     def enable_graph_capture(self, enabled: bool = True):
@@ -190,12 +202,14 @@ class DiffusionSampler:
             device = (next(self.model.parameters()).device
                       if hasattr(self.model, "parameters") else "cpu")
         if start_step is None:
-            start_step = self.noise_schedule.max_timesteps
-            if isinstance(start_step, float):
-                start_step = 1000 if start_step <= 1 else int(start_step)
-        if isinstance(self.noise_schedule.max_timesteps, (int, float)) and \
-                self.noise_schedule.max_timesteps <= 1:
-            start_step = 1000 if start_step is None else start_step
+            # Callers operate in the 1000-step convention; scale_steps maps to
+            # the scheduler range (reference samplers/common.py:178-181 — the
+            # reference README's inference passes start_step=1000 explicitly).
+            # Continuous schedulers report max_timesteps == 1, so using it as
+            # the step range would collapse sampling to a single model eval.
+            mt = self.noise_schedule.max_timesteps
+            start_step = 1000 if (isinstance(mt, (int, float)) and mt <= 1) \
+                else int(mt)
 
         if priors is None:
             rngstate, key = rngstate.get_random_key()
@@ -227,6 +241,11 @@ class DiffusionSampler:
                     or not self._graphed.matches(samples, model_conditioning_inputs)):
                 self._graphed = _GraphedSampleModel(self, samples,
                                                    model_conditioning_inputs)
+            else:
+                # Same shapes as the captured graph, possibly different
+                # conditioning contents (new prompts on a cached sampler):
+                # refresh the static conditioning buffers before replaying.
+                self._graphed.update_conds(model_conditioning_inputs)
 
             def sample_model_fn(x_t, t, *cond):
                 return self._graphed(x_t, t)
@@ -285,9 +304,10 @@ class _GraphedSampleModel:
     """hipGraph-captured sample_model: static input/cond/output buffers, one
     graph replay per model evaluation (SURVEY.md §2.5 MI355X plan).
 
-    Conditioning tensors are copied into static buffers ONCE at capture (they
-    are constant across the sampling loop); x_t and t stream through static
-    buffers each call.
+    Conditioning tensors live in static buffers refreshed via update_conds()
+    at the start of every generate_samples call (constant across the sampling
+    loop, so one copy per call is capture-safe); x_t and t stream through
+    static buffers each call.
     """
 
     def __init__(self, sampler: DiffusionSampler, x: torch.Tensor,
@@ -316,11 +336,19 @@ class _GraphedSampleModel:
                 self.static_x, self.static_t, *self.static_conds)
 
     def matches(self, x: torch.Tensor, conds) -> bool:
-        return tuple(x.shape) == self.shape and x.dtype == self.dtype and \
-            len(conds) == len(self.static_conds)
+        return (tuple(x.shape) == self.shape and x.dtype == self.dtype
+                and len(conds) == len(self.static_conds)
+                and all(tuple(c.shape) == tuple(s.shape) and c.dtype == s.dtype
+                        for c, s in zip(conds, self.static_conds)))
+
+    def update_conds(self, conds):
+        for s, c in zip(self.static_conds, conds):
+            s.copy_(c)
 
     def __call__(self, x_t: torch.Tensor, t: torch.Tensor):
         self.static_x.copy_(x_t)
         self.static_t.copy_(t)
         self.graph.replay()
-        return self.out_x0.clone(), self.out_eps.clone(), self.out_raw
+        # out_raw is cloned too: the next replay overwrites the static output
+        # in place, so an unprotected reference would be silently corrupted.
+        return self.out_x0.clone(), self.out_eps.clone(), self.out_raw.clone()

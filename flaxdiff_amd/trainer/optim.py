@@ -34,7 +34,8 @@ class FlatAdamWEMA:
         self.lr_schedule = lr_schedule
         self.grad_clip_norm = grad_clip_norm
         self.skip_nonfinite = skip_nonfinite
-        self.skipped_steps = 0
+        self._skipped_host = 0
+        self._skip_dev: Optional[torch.Tensor] = None  # GPU-path skip counter
         self.step_count = 0
 
         params = [p for p in module.parameters() if p.requires_grad]
@@ -79,6 +80,13 @@ class FlatAdamWEMA:
             return self.lr_schedule(self.step_count)
         return self.lr
 
+    @property
+    def skipped_steps(self) -> int:
+        n = self._skipped_host
+        if self._skip_dev is not None:
+            n += int(self._skip_dev.item())
+        return n
+
     @torch.no_grad()
     def step(self, grad_scale: float = 1.0):
         """grad_scale multiplies gradients before use (e.g. 1/world_size so the
@@ -86,27 +94,46 @@ class FlatAdamWEMA:
         self.step_count += 1
         lr = self.current_lr()
 
+        if self.flat.is_cuda and ops.hip_available():
+            # Clip factor and nonfinite gate are built ON DEVICE so the hot
+            # step never does a GPU->host sync. A gated (skipped)
+            # step leaves step_count advanced on the host — bias correction
+            # deviates by one exponent tick per (rare, abnormal) skip.
+            scale_dev = None
+            if self.grad_clip_norm is not None or self.skip_nonfinite:
+                gnorm = self.flat_grad.norm() * grad_scale
+                scale_dev = gnorm.new_full((1,), grad_scale)
+                if self.grad_clip_norm is not None:
+                    scale_dev = scale_dev * torch.clamp(
+                        self.grad_clip_norm / (gnorm + 1e-6), max=1.0)
+                if self.skip_nonfinite:
+                    if self._skip_dev is None:
+                        self._skip_dev = torch.zeros(1, dtype=torch.int32,
+                                                     device=self.flat.device)
+                    scale_dev = torch.where(torch.isfinite(gnorm), scale_dev,
+                                            torch.zeros_like(scale_dev))
+                scale_dev = scale_dev.reshape(1).float().contiguous()
+            ops.fused_adamw_ema(self.flat, self.flat_grad, self.exp_avg,
+                                self.exp_avg_sq, self.ema, self.flat_bf16,
+                                lr=lr, beta1=self.beta1, beta2=self.beta2,
+                                eps=self.eps, weight_decay=self.weight_decay,
+                                step=self.step_count, ema_decay=self.ema_decay,
+                                grad_scale=grad_scale, scale_dev=scale_dev,
+                                skip_ctr=self._skip_dev if self.skip_nonfinite else None)
+            return
+
         if self.grad_clip_norm is not None or self.skip_nonfinite:
             gnorm = float(self.flat_grad.norm()) * grad_scale
             if self.skip_nonfinite and not math.isfinite(gnorm):
                 # DynamicScale skip-on-nonfinite parity (reference
                 # diffusion_trainer.py:229-240): drop the step, keep state.
                 self.step_count -= 1
-                self.skipped_steps += 1
+                self._skipped_host += 1
                 return
             if self.grad_clip_norm is not None:
                 clip = self.grad_clip_norm / (gnorm + 1e-6)
                 if clip < 1.0:
                     grad_scale = grad_scale * clip
-
-        if self.flat.is_cuda and ops.hip_available():
-            ops.fused_adamw_ema(self.flat, self.flat_grad, self.exp_avg,
-                                self.exp_avg_sq, self.ema, self.flat_bf16,
-                                lr=lr, beta1=self.beta1, beta2=self.beta2,
-                                eps=self.eps, weight_decay=self.weight_decay,
-                                step=self.step_count, ema_decay=self.ema_decay,
-                                grad_scale=grad_scale)
-            return
 
         g = self.flat_grad
         if grad_scale != 1.0:
