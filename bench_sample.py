@@ -45,9 +45,14 @@ def main():
     null_ctx = torch.zeros(1, 77, 768, device=dev)
 
     def bench(sampler_cls, graph: bool, label: str):
-        sampler = sampler_cls(model=lambda x, t, *c: model(
-                                  x.to(dtype), t,
-                                  null_ctx.expand(x.shape[0], -1, -1).to(dtype)).float(),
+        calls = {"n": 0}
+
+        def wrapped_model(x, t, *c):
+            calls["n"] += 1  # counts evals (graph path: captures, not replays)
+            return model(x.to(dtype), t,
+                         null_ctx.expand(x.shape[0], -1, -1).to(dtype)).float()
+
+        sampler = sampler_cls(model=wrapped_model,
                               noise_schedule=schedule,
                               model_output_transform=transform,
                               guidance_scale=args.guidance,
@@ -57,6 +62,13 @@ def main():
         kw = dict(num_samples=args.batch, resolution=args.resolution,
                   diffusion_steps=args.steps, device=dev, dtype=torch.float32)
         sampler.generate_samples(rngstate=RandomMarkovState(1), **kw)  # warmup
+        if not graph:
+            # NFE self-check: the warmup batch must have evaluated the model
+            # >= diffusion_steps times (Heun: 2/step). Guards against the
+            # round-1 regression where continuous schedulers collapsed to 1.
+            nfe_min = args.steps
+            assert calls["n"] >= nfe_min, \
+                f"only {calls['n']} model evals for {args.steps}-step sampling"
         if use_gpu:
             torch.cuda.synchronize()
         t0 = time.perf_counter()
