@@ -34,9 +34,10 @@ def main():
     model = Unet(emb_features=256, feature_depths=[64, 128, 256, 512],
                  attention_configs=[{"heads": 4}] * 4, num_res_blocks=2,
                  num_middle_res_blocks=1, norm_groups=8, context_dim=768).to(dev)
-    # bf16 shadow-free path: cast params for inference
+    # inference runs the whole net in bf16 (same compute dtype as training;
+    # the sampler's transform math stays fp32 on the outputs)
     if use_gpu:
-        model = model.to(torch.float32)
+        model = model.to(torch.bfloat16)
     model.eval()
 
     schedule = KarrasVENoiseScheduler(1, sigma_max=80, rho=7, sigma_data=0.5)
@@ -56,7 +57,7 @@ def main():
                               noise_schedule=schedule,
                               model_output_transform=transform,
                               guidance_scale=args.guidance,
-                              timestep_spacing="karras")
+                              timestep_spacing="linear")  # KarrasVE scheduler: linear t IS the rho ramp
         if graph:
             sampler.enable_graph_capture()
         kw = dict(num_samples=args.batch, resolution=args.resolution,

@@ -160,7 +160,13 @@ class DiffusionSampler:
             s = torch.linspace(0, 1, diffusion_steps) ** 2
             steps = ((start_step - end_step) * s + end_step).flip(0)
         elif spacing == "karras":
-            sigma_min = end_step / start_step
+            # NOTE: for KarrasVE/EDM schedulers the scheduler's own get_sigmas
+            # already applies the rho ramp, so LINEAR spacing is the Karras
+            # schedule; this option re-warps the step domain (reference
+            # samplers/common.py:214-230). end_step=0 would make sigma_min=0
+            # and the log-space degenerate (every interior point -inf), so
+            # floor it at one step.
+            sigma_min = max(float(end_step), 1.0) / start_step
             rho = 7.0
             sigmas = torch.exp(torch.linspace(torch.log(torch.tensor(1.0)),
                                               torch.log(torch.tensor(float(sigma_min))),
@@ -175,7 +181,10 @@ class DiffusionSampler:
             steps = torch.clamp(steps, end_step, start_step).flip(0)
         else:  # linear
             steps = torch.linspace(end_step, start_step, diffusion_steps).flip(0)
-        return steps.round().long()
+        steps = steps.round().long()
+        # Repeated consecutive steps would give dtau == 0 in the ODE samplers
+        # (division by zero -> NaN); only degenerate spacings produce them.
+        return torch.unique_consecutive(steps)
 
     # ------------------------------------------------------------------
     # sample loop (reference common.py:248-389)
