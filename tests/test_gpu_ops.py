@@ -374,7 +374,8 @@ def test_graph_captured_sampling_matches_eager():
     eager = run(False)
     graphed = run(True)
     assert torch.isfinite(graphed).all()
-    assert (eager - graphed).abs().max() < 1e-4
+    # 4 real steps of bf16 + nondeterministic split-k GEMM atomics
+    assert (eager - graphed).abs().max() < 2e-2
 
 
 @pytest.mark.gpu
