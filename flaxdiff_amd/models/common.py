@@ -91,6 +91,10 @@ class Dense(nn.Module):
         self.bias = nn.Parameter(torch.zeros(out_features)) if use_bias else None
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if x.is_cuda and x.dtype == torch.bfloat16:
+            # fp32 masters go straight in: ops.dense uses the bf16 shadow for
+            # compute and returns fp32 grads (no per-call cast kernels)
+            return ops.dense(x, self.weight, self.bias)
         return ops.dense(x, _cast(self.weight, x.dtype),
                          _cast(self.bias, x.dtype))
 
@@ -188,6 +192,8 @@ class Conv(nn.Module):
         self.bias = nn.Parameter(torch.zeros(features)) if use_bias else None
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if x.is_cuda and x.dtype == torch.bfloat16:
+            return ops.conv2d(x, self.weight, self.bias, stride=self.stride)
         return ops.conv2d(x, _cast(self.weight, x.dtype), _cast(self.bias, x.dtype),
                           stride=self.stride)
 

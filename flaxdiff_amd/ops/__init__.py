@@ -162,30 +162,54 @@ def _conv2d_wgrad_gemms(dy: torch.Tensor, x: torch.Tensor, kh: int, kw: int,
     return dw
 
 
+def _kernel_view(p: torch.Tensor) -> torch.Tensor:
+    """bf16 tensor the compute kernels consume. An fp32 master param with a
+    bf16 shadow (trainer/optim.py) uses the shadow — no cast kernel; the
+    grads then flow back in fp32 straight into the flat grad buffer (no
+    fp32->bf16->fp32 round trip through a _ShadowCast node)."""
+    if p.dtype == torch.bfloat16:
+        return p
+    sh = getattr(p, "_shadow_bf16", None)
+    if sh is not None:
+        return sh
+    return p.to(torch.bfloat16)
+
+
+def _routable_param(p: Optional[torch.Tensor]) -> bool:
+    return p is None or p.dtype == torch.bfloat16 or \
+        getattr(p, "_shadow_bf16", None) is not None
+
+
 class _Conv2dFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, w, b, stride):
         ext = _require_ext()
-        y = ext.conv2d_fwd(x, w, b if b is not None else torch.Tensor(), stride)
-        ctx.save_for_backward(x, w)
+        wk = _kernel_view(w)
+        # fp32 bias feeds the kernel's fp32 epilogue directly (no cast)
+        y = ext.conv2d_fwd(x, wk, b if b is not None else torch.Tensor(), stride)
+        ctx.save_for_backward(x, wk)
         ctx.stride = stride
         ctx.has_bias = b is not None
+        ctx.w_dtype = w.dtype
+        ctx.b_dtype = b.dtype if b is not None else None
         return y
 
     @staticmethod
     def backward(ctx, dy):
-        x, w = ctx.saved_tensors
+        x, wk = ctx.saved_tensors
         ext = _require_ext()
         dy = dy.contiguous()
-        dx = ext.conv2d_dgrad(dy, w, ctx.stride, x.shape[1], x.shape[2]) \
+        dx = ext.conv2d_dgrad(dy, wk, ctx.stride, x.shape[1], x.shape[2]) \
             if ctx.needs_input_grad[0] else None
         dw = db = None
         if ctx.needs_input_grad[1]:
-            dw, _db_unused = ext.conv2d_wgrad(dy, x, w.shape[0], w.shape[1],
+            dw, _db_unused = ext.conv2d_wgrad(dy, x, wk.shape[0], wk.shape[1],
                                               ctx.stride)
-            dw = dw.to(w.dtype)
+            dw = dw if ctx.w_dtype == torch.float32 else dw.to(ctx.w_dtype)
         if ctx.has_bias:
-            db = _bias_grad(dy).to(w.dtype)
+            db = _bias_grad(dy)          # fp32
+            if ctx.b_dtype != torch.float32:
+                db = db.to(ctx.b_dtype)
         return dx, dw, db, None
 
 
@@ -198,40 +222,52 @@ class _DenseFn(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x2d, w2d, b):
-        y = torch.matmul(x2d, w2d)
+        wk = _kernel_view(w2d)
+        y = torch.matmul(x2d, wk)
         if b is not None:
-            y = y + b
-        ctx.save_for_backward(x2d, w2d)
+            y = y + _kernel_view(b)
+        ctx.save_for_backward(x2d, wk)
         ctx.has_bias = b is not None
+        ctx.w_dtype = w2d.dtype
+        ctx.b_dtype = b.dtype if b is not None else None
         return y
 
     @staticmethod
     def backward(ctx, dy):
-        x2d, w2d = ctx.saved_tensors
+        x2d, wk = ctx.saved_tensors
         dy = dy.contiguous()
         ext = _require_ext()
-        dx = torch.matmul(dy, w2d.t()) if ctx.needs_input_grad[0] else None
+        dx = torch.matmul(dy, wk.t()) if ctx.needs_input_grad[0] else None
         dw = None
         if ctx.needs_input_grad[1]:
             M = x2d.shape[0]
             dw4, _ = ext.conv2d_wgrad(dy.view(M, 1, 1, dy.shape[1]),
                                       x2d.view(M, 1, 1, x2d.shape[1]), 1, 1, 1)
-            dw = dw4.view(x2d.shape[1], dy.shape[1]).to(w2d.dtype)
-        db = _bias_grad(dy).to(dy.dtype) if ctx.has_bias else None
+            dw = dw4.view(x2d.shape[1], dy.shape[1])
+            if ctx.w_dtype != torch.float32:
+                dw = dw.to(ctx.w_dtype)
+        db = None
+        if ctx.has_bias:
+            db = _bias_grad(dy)
+            if ctx.b_dtype != torch.float32:
+                db = db.to(ctx.b_dtype)
         return dx, dw, db
 
 
 def dense(x: torch.Tensor, w: torch.Tensor, b: Optional[torch.Tensor] = None
           ) -> torch.Tensor:
     """x [..., Cin] @ w [Cin, Cout] (+ b): library GEMM forward, split-M
-    wgrad backward on GPU."""
-    if _use_hip(x) and x.dtype == torch.bfloat16 and w.dtype == torch.bfloat16:
+    wgrad backward on GPU. w/b may be fp32 masters with bf16 shadows."""
+    if _use_hip(x) and x.dtype == torch.bfloat16 and _routable_param(w) \
+            and _routable_param(b):
         lead = x.shape[:-1]
         y = _DenseFn.apply(x.reshape(-1, x.shape[-1]).contiguous(), w, b)
         return y.reshape(*lead, w.shape[1])
+    if w.dtype != x.dtype:
+        w = w.to(x.dtype)
     y = torch.matmul(x, w)
     if b is not None:
-        y = y + b
+        y = y + b.to(y.dtype)
     return y
 
 
@@ -246,15 +282,20 @@ def _bias_grad(dy: torch.Tensor) -> torch.Tensor:
 
 def conv2d(x: torch.Tensor, w: torch.Tensor, b: Optional[torch.Tensor] = None,
            stride: int = 1) -> torch.Tensor:
-    """SAME-padded NHWC conv. 1x1 convs route to a plain GEMM."""
+    """SAME-padded NHWC conv. 1x1 convs route to a plain GEMM.
+    w/b may be fp32 master params with bf16 shadows (no cast kernels)."""
     kh, kw = w.shape[0], w.shape[1]
     if kh == 1 and kw == 1 and stride == 1:
-        y = dense(x.reshape(-1, w.shape[2]).to(w.dtype),
+        y = dense(x.reshape(-1, w.shape[2]),
                   w.reshape(w.shape[2], w.shape[3]), b)
         return y.reshape(*x.shape[:-1], w.shape[3])
-    if _use_hip(x):
+    if _use_hip(x) and x.dtype == torch.bfloat16 and _routable_param(w):
         return _Conv2dFn.apply(x.contiguous(), w.contiguous(),
                                b.contiguous() if b is not None else None, stride)
+    if w.dtype != x.dtype:
+        w = w.to(x.dtype)
+    if b is not None and b.dtype != x.dtype:
+        b = b.to(x.dtype)
     return reference.conv2d_nhwc(x, w, b, stride=stride, padding="same")
 
 
