@@ -286,8 +286,14 @@ def conv2d(x: torch.Tensor, w: torch.Tensor, b: Optional[torch.Tensor] = None,
     w/b may be fp32 master params with bf16 shadows (no cast kernels)."""
     kh, kw = w.shape[0], w.shape[1]
     if kh == 1 and kw == 1 and stride == 1:
-        y = dense(x.reshape(-1, w.shape[2]),
-                  w.reshape(w.shape[2], w.shape[3]), b)
+        w2 = w.reshape(w.shape[2], w.shape[3])
+        # reshape drops tensor attributes: re-attach the bf16 shadow so the
+        # fp32 master still routes through _DenseFn (grads flow through the
+        # reshape view back to the param)
+        sh = getattr(w, "_shadow_bf16", None)
+        if sh is not None and w.dtype != torch.bfloat16:
+            w2._shadow_bf16 = sh.reshape(w.shape[2], w.shape[3])
+        y = dense(x.reshape(-1, w.shape[2]), w2, b)
         return y.reshape(*x.shape[:-1], w.shape[3])
     if _use_hip(x) and x.dtype == torch.bfloat16 and _routable_param(w):
         return _Conv2dFn.apply(x.contiguous(), w.contiguous(),
