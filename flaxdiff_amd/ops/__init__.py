@@ -414,8 +414,9 @@ class _AttentionFn(torch.autograd.Function):
         # attn_bwd_ab.py). Larger head dims fall through to the GEMMs.
         if k.shape[2] <= 128 and q.shape[3] <= 32:
             ext = _require_ext()
-            dq, dk, dv = ext.attn_bwd_smallkv(q, k, v, do.contiguous(), lse,
-                                              scale)
+            fn = ext.attn_bwd_smallkv if os.environ.get("FD_ATTN_BWD_V1") \
+                else ext.attn_bwd_smallkv_v2
+            dq, dk, dv = fn(q, k, v, do.contiguous(), lse, scale)
             return dq, dk, dv, None
         # General shapes: recompute P row-exactly from the saved log-sum-exp,
         # then the grads are plain batched GEMMs (library GEMM on MFMA).
