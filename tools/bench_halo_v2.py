@@ -63,6 +63,7 @@ def main():
         (256, 64, 64, 64),
         (256, 32, 128, 128),
         (256, 16, 256, 256),
+        (256, 8, 512, 512),
     ]):
         run(B, HW, Ci, Co, check=(i < 4))
 
