@@ -70,13 +70,14 @@ def main():
         (8, 32, 32, 128, 128),
         (8, 16, 16, 256, 256),
         (8, 16, 16, 512, 512),
+        (8, 8, 8, 512, 512),
         (256, 64, 64, 64, 64),     # bench level-0 shape
         (256, 32, 32, 128, 128),   # bench level-1
         (256, 16, 16, 256, 256),   # level-2
         (256, 8, 8, 512, 512),     # level-3 (W=8: falls back to old kernel)
     ]
     for i, (B, H, W, Ci, Co) in enumerate(shapes):
-        run(B, H, W, Ci, Co, check=(i < 4))
+        run(B, H, W, Ci, Co, check=(i < 5))
 
 
 if __name__ == "__main__":
