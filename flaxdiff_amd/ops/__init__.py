@@ -56,6 +56,7 @@ def _require_ext():
 
 
 _FORCE_TORCH = os.environ.get("FLAXDIFF_FORCE_TORCH_OPS", "0") == "1"
+_GEMM_LIB = os.environ.get("FD_GEMM_LIB", "0") == "1"  # revert dense GEMMs to hipBLASLt
 
 
 def _use_hip(x: torch.Tensor) -> bool:
@@ -223,9 +224,18 @@ class _DenseFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x2d, w2d, b):
         wk = _kernel_view(w2d)
-        y = torch.matmul(x2d, wk)
-        if b is not None:
-            y = y + _kernel_view(b)
+        if not _GEMM_LIB and x2d.shape[1] % 64 == 0:
+            # hand-written MFMA GEMM (fused fp32 bias) — no hipBLASLt on the
+            # declared hot path (BASELINE.json north star)
+            ext = _require_ext()
+            bf = b if b is not None else torch.Tensor()
+            if bf.numel() and bf.dtype != torch.float32:
+                bf = bf.float()
+            y = ext.gemm_fwd(x2d, wk, bf)
+        else:
+            y = torch.matmul(x2d, wk)
+            if b is not None:
+                y = y + _kernel_view(b)
         ctx.save_for_backward(x2d, wk)
         ctx.has_bias = b is not None
         ctx.w_dtype = w2d.dtype
@@ -237,7 +247,13 @@ class _DenseFn(torch.autograd.Function):
         x2d, wk = ctx.saved_tensors
         dy = dy.contiguous()
         ext = _require_ext()
-        dx = torch.matmul(dy, wk.t()) if ctx.needs_input_grad[0] else None
+        if ctx.needs_input_grad[0]:
+            if not _GEMM_LIB and dy.shape[1] % 64 == 0:
+                dx = ext.gemm_dx(dy, wk)
+            else:
+                dx = torch.matmul(dy, wk.t())
+        else:
+            dx = None
         dw = None
         if ctx.needs_input_grad[1]:
             M = x2d.shape[0]
