@@ -176,6 +176,20 @@ def _kernel_view(p: torch.Tensor) -> torch.Tensor:
     return p.to(torch.bfloat16)
 
 
+def _kernel_view_t(p: torch.Tensor) -> torch.Tensor:
+    """[N,K] transposed bf16 view of a [K,N] dense weight for the NT GEMM.
+    Optimizer-managed params carry a per-step-refreshed transposed shadow;
+    otherwise (inference) transpose on the fly."""
+    t = getattr(p, "_shadow_bf16_t", None)
+    if t is not None:
+        return t
+    return _kernel_view(p).t().contiguous()
+
+
+def transpose_shadows(src_flat, dst_flat, tiles):
+    _require_ext().transpose_shadows(src_flat, dst_flat, tiles)
+
+
 def _routable_param(p: Optional[torch.Tensor]) -> bool:
     return p is None or p.dtype == torch.bfloat16 or \
         getattr(p, "_shadow_bf16", None) is not None
@@ -226,12 +240,13 @@ class _DenseFn(torch.autograd.Function):
         wk = _kernel_view(w2d)
         if not _GEMM_LIB and x2d.shape[1] % 64 == 0:
             # hand-written MFMA GEMM (fused fp32 bias) — no hipBLASLt on the
-            # declared hot path (BASELINE.json north star)
+            # declared hot path (BASELINE.json north star). The weight comes
+            # in transposed ([N,K] NT layout, the fast A-row read).
             ext = _require_ext()
             bf = b if b is not None else torch.Tensor()
             if bf.numel() and bf.dtype != torch.float32:
                 bf = bf.float()
-            y = ext.gemm_fwd(x2d, wk, bf)
+            y = ext.gemm_nt(x2d, _kernel_view_t(w2d), bf)
         else:
             y = torch.matmul(x2d, wk)
             if b is not None:
@@ -309,6 +324,9 @@ def conv2d(x: torch.Tensor, w: torch.Tensor, b: Optional[torch.Tensor] = None,
         sh = getattr(w, "_shadow_bf16", None)
         if sh is not None and w.dtype != torch.bfloat16:
             w2._shadow_bf16 = sh.reshape(w.shape[2], w.shape[3])
+            sht = getattr(w, "_shadow_bf16_t", None)
+            if sht is not None:
+                w2._shadow_bf16_t = sht
         y = dense(x.reshape(-1, w.shape[2]), w2, b)
         return y.reshape(*x.shape[:-1], w.shape[3])
     if _use_hip(x) and x.dtype == torch.bfloat16 and _routable_param(w):

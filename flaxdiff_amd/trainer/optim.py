@@ -71,6 +71,47 @@ class FlatAdamWEMA:
         self.flat_bf16.copy_(self.flat)
         self.total = total
 
+        # Transposed bf16 shadows for 64-divisible 2-D dense weights: the
+        # MFMA GEMM forward consumes W^T [N,K] (fast NT operand layout); one
+        # batched 64x64 transpose kernel refreshes them per step instead of
+        # a per-call transpose (ops/hip/gemm_bf16.hip transpose_shadows).
+        self.flat_bf16_t = None
+        self._t_tiles = None
+        def _t_shape(p):
+            if p.dim() == 2:
+                kn = (p.shape[0], p.shape[1])
+            elif p.dim() == 4 and p.shape[0] == 1 and p.shape[1] == 1:
+                kn = (p.shape[2], p.shape[3])     # 1x1 conv weight
+            else:
+                return None
+            return kn if kn[0] % 64 == 0 and kn[1] % 64 == 0 else None
+
+        t_params = [(i, p, _t_shape(p)) for i, p in enumerate(params)
+                    if _t_shape(p) is not None]
+        if t_params and device.type == "cuda":
+            tot_t = sum(p.numel() for _, p, _kn in t_params)
+            self.flat_bf16_t = torch.zeros(tot_t, dtype=torch.bfloat16,
+                                           device=device)
+            tiles = []
+            off_t = 0
+            for i, p, (K, N) in t_params:
+                src = self.offsets[i]
+                for n0 in range(0, N, 64):
+                    for k0 in range(0, K, 64):
+                        tiles.append((src + k0 * N + n0, off_t + n0 * K + k0,
+                                      N, K))
+                p._shadow_bf16_t = self.flat_bf16_t[off_t:off_t + p.numel()] \
+                    .view(N, K)
+                off_t += p.numel()
+            self._t_tiles = torch.tensor(tiles, dtype=torch.int32,
+                                         device=device)
+            self._refresh_t()
+
+    def _refresh_t(self):
+        if self.flat_bf16_t is not None and ops.hip_available():
+            ops.transpose_shadows(self.flat_bf16, self.flat_bf16_t,
+                                  self._t_tiles)
+
     # ------------------------------------------------------------------
     def zero_grad(self):
         self.flat_grad.zero_()
@@ -120,6 +161,7 @@ class FlatAdamWEMA:
                                 step=self.step_count, ema_decay=self.ema_decay,
                                 grad_scale=grad_scale, scale_dev=scale_dev,
                                 skip_ctr=self._skip_dev if self.skip_nonfinite else None)
+            self._refresh_t()
             return
 
         if self.grad_clip_norm is not None or self.skip_nonfinite:
@@ -162,11 +204,13 @@ class FlatAdamWEMA:
         saved = self.flat.clone()
         self.flat.copy_(self.ema)
         self.flat_bf16.copy_(self.flat)
+        self._refresh_t()
         return saved
 
     def restore_params(self, saved: torch.Tensor):
         self.flat.copy_(saved)
         self.flat_bf16.copy_(self.flat)
+        self._refresh_t()
 
     # ------------------------------------------------------------------
     def state_dict(self):
@@ -186,6 +230,7 @@ class FlatAdamWEMA:
             self.exp_avg_sq.copy_(sd["exp_avg_sq"])
             self.ema.copy_(sd["ema"])
             self.flat_bf16.copy_(self.flat)
+        self._refresh_t()
         self.step_count = int(sd["step_count"])
 
 
