@@ -46,15 +46,21 @@ def main():
     from flaxdiff_amd.trainer import DiffusionTrainer
 
     torch.manual_seed(1234)
+    # FD_BENCH_TINY: plumbing dry-run only (CI launches the full torchrun
+    # rendezvous + DP path on CPU with a toy model; never a perf number)
+    tiny = os.environ.get("FD_BENCH_TINY") == "1"
+    if tiny:
+        args.global_batch = min(args.global_batch, 4 * n_gpus)
+        args.resolution = 16
     model = Unet(
         output_channels=3,
-        emb_features=256,
-        feature_depths=[64, 128, 256, 512],
-        attention_configs=[{"heads": 4}] * 4,
-        num_res_blocks=2,
+        emb_features=32 if tiny else 256,
+        feature_depths=[8, 16] if tiny else [64, 128, 256, 512],
+        attention_configs=[{"heads": 2}] * (2 if tiny else 4),
+        num_res_blocks=1 if tiny else 2,
         num_middle_res_blocks=1,
-        norm_groups=8,
-        context_dim=768,
+        norm_groups=4 if tiny else 8,
+        context_dim=16 if tiny else 768,
     )
     trainer = DiffusionTrainer(
         model,
@@ -65,6 +71,7 @@ def main():
         compute_dtype=compute_dtype,
         distributed=(world > 1),
         optimizer_kwargs={"lr": 2.7e-4},
+        **({"text_context_shape": (4, 16)} if tiny else {}),
     )
     dev = trainer.device
 

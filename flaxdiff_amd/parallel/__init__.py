@@ -116,7 +116,13 @@ class GradBucketSynchronizer:
     """
 
     def __init__(self, params: List[torch.nn.Parameter], flat_grad: torch.Tensor,
-                 offsets: List[int], bucket_bytes: int = 64 << 20):
+                 offsets: List[int], bucket_bytes: Optional[int] = None):
+        if bucket_bytes is None:
+            # xGMI tuning knob (docs/multigpu_tuning.md): ring all-reduce on
+            # 8 GPUs is per-link bound (7 x ~153 GB/s point-to-point), so the
+            # bucket must be big enough to amortize per-step latency but
+            # small enough to start reducing early in backward.
+            bucket_bytes = int(os.environ.get("FD_BUCKET_MB", "64")) << 20
         self.flat_grad = flat_grad
         self.enabled = dist.is_initialized()
         elem = flat_grad.element_size()

@@ -217,3 +217,71 @@ def test_data_parallel_resume(tmp_path):
         for r in range(world):
             assert results[r]["step"] == 3
             assert results[r]["params_equal"]
+
+
+@pytest.mark.timeout(420)
+def test_data_parallel_four_ranks():
+    """4-rank gloo lockstep (VERDICT r1 #5: beyond world_size=2)."""
+    world = 4
+    ctx = mp.get_context("spawn")
+    with ctx.Manager() as man:
+        results = man.dict()
+        procs = [ctx.Process(target=_worker, args=(r, world, 29741, results))
+                 for r in range(world)]
+        for p in procs:
+            p.start()
+        for p in procs:
+            p.join(timeout=360)
+        for p in procs:
+            assert p.exitcode == 0
+        res = dict(results)
+    assert all(res[r]["params_equal"] for r in range(world))
+    for r in range(1, world):
+        assert res[0]["losses"] == pytest.approx(res[r]["losses"], rel=1e-5)
+
+
+@pytest.mark.timeout(600)
+@pytest.mark.slow
+def test_data_parallel_eight_ranks():
+    """8-rank gloo lockstep — the driver's scaling-run world size."""
+    world = 8
+    ctx = mp.get_context("spawn")
+    with ctx.Manager() as man:
+        results = man.dict()
+        procs = [ctx.Process(target=_worker, args=(r, world, 29751, results))
+                 for r in range(world)]
+        for p in procs:
+            p.start()
+        for p in procs:
+            p.join(timeout=540)
+        for p in procs:
+            assert p.exitcode == 0
+        res = dict(results)
+    assert all(res[r]["params_equal"] for r in range(world))
+
+
+@pytest.mark.timeout(420)
+def test_bench_dry_run_torchrun_cpu():
+    """bench.py under the driver's exact torchrun launch, 4 CPU ranks.
+
+    FD_BENCH_TINY=1 shrinks the model so this is a PLUMBING test of the
+    rendezvous + DP + JSON contract, not a perf number."""
+    import json
+    import subprocess
+    import sys
+
+    env = dict(os.environ)
+    env.update({"FD_BENCH_TINY": "1", "MASTER_ADDR": "127.0.0.1"})
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+           "--nproc-per-node", "4", "--master-addr", "127.0.0.1",
+           "--master-port", "29761", os.path.join(repo, "bench.py"),
+           "--gpus", "4", "--steps", "2", "--warmup", "1"]
+    out = subprocess.run(cmd, capture_output=True, text=True, timeout=360,
+                         env=env, cwd=repo)
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = [l for l in out.stdout.splitlines() if l.startswith("{")][-1]
+    rec = json.loads(line)
+    assert rec["n_gpus"] == 4
+    assert rec["steps"] == 2
+    assert rec["value"] > 0
