@@ -100,10 +100,14 @@ class CLIPTextEncoder(ConditioningEncoder):
 
     @torch.no_grad()
     def encode_from_tokens(self, tokens):
-        if isinstance(tokens, dict):
-            tokens = {k: torch.as_tensor(v).to(self.device) for k, v in tokens.items()}
-            out = self.model(input_ids=tokens["input_ids"],
-                             attention_mask=tokens.get("attention_mask"))
+        # transformers tokenizers return a BatchEncoding (a UserDict, NOT a
+        # dict subclass) — treat any mapping with input_ids as the dict form
+        if hasattr(tokens, "keys") and "input_ids" in tokens:
+            ids = torch.as_tensor(tokens["input_ids"]).to(self.device)
+            mask = tokens.get("attention_mask")
+            if mask is not None:
+                mask = torch.as_tensor(mask).to(self.device)
+            out = self.model(input_ids=ids, attention_mask=mask)
         else:
             out = self.model(input_ids=torch.as_tensor(tokens).to(self.device))
         return out.last_hidden_state
