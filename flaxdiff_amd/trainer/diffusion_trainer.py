@@ -73,17 +73,25 @@ class DiffusionTrainer(SimpleTrainer):
 
         B = images.shape[0]
 
-        # conditioning context
-        text = batch.get("text_emb")
-        if text is None:
-            text = self.null_context.unsqueeze(0).expand(B, *self.null_context.shape)
+        # conditioning context: either N pre-processed modalities
+        # ("cond_embs", CFG dropout already applied — GeneralDiffusionTrainer)
+        # or the single-text path with bernoulli null splice here
+        conds = batch.get("cond_embs")
+        if conds is not None:
+            conds = tuple(torch.as_tensor(c).to(dev, self.compute_dtype)
+                          for c in conds)
         else:
-            text = text.to(dev, self.compute_dtype)
-            # CFG dropout: bernoulli null splice (reference :181-190)
-            self.rngs, key = self.rngs.get_random_key()
-            mask = key.bernoulli((B,), self.unconditional_prob, device=dev)
-            text = torch.where(mask[:, None, None],
-                               self.null_context.unsqueeze(0).to(text.dtype), text)
+            text = batch.get("text_emb")
+            if text is None:
+                text = self.null_context.unsqueeze(0).expand(B, *self.null_context.shape)
+            else:
+                text = text.to(dev, self.compute_dtype)
+                # CFG dropout: bernoulli null splice (reference :181-190)
+                self.rngs, key = self.rngs.get_random_key()
+                mask = key.bernoulli((B,), self.unconditional_prob, device=dev)
+                text = torch.where(mask[:, None, None],
+                                   self.null_context.unsqueeze(0).to(text.dtype), text)
+            conds = (text,)
 
         # timesteps + noise (reference :192-195)
         timesteps, self.rngs = self.noise_schedule.generate_timesteps(B, self.rngs, device=dev)
@@ -101,7 +109,7 @@ class DiffusionTrainer(SimpleTrainer):
             t_in = t_in.to(dev)
 
         self.optimizer.zero_grad()
-        pred = self.model(x_in, t_in, text)
+        pred = self.model(x_in, t_in, *conds)
         pred = self.model_output_transform.pred_transform(x_t, pred, rates)
 
         weights = self.noise_schedule.get_weights(timesteps, get_coeff_shapes_tuple(images)).to(dev)

@@ -53,14 +53,21 @@ class GeneralDiffusionTrainer(DiffusionTrainer):
             if sample_key in batch and "image" not in batch:
                 batch = dict(batch)
                 batch["image"] = batch[sample_key]
-            # process conditions -> a single context tensor for now (text)
-            if self.input_config.conditions and "text_emb" not in batch:
-                cond = self.input_config.conditions[0]
-                key = cond.encoder.key if hasattr(cond.encoder, "key") else None
-                if key is not None and key in batch:
+            # N-modality conditioning: CFG-dropout mask shared across
+            # modalities, each replaced by its own null embedding
+            # (reference inputs/__init__.py:123-146 process_conditioning)
+            if self.input_config.conditions and "cond_embs" not in batch \
+                    and "text_emb" not in batch:
+                keys = [c.conditioning_data_key or c.encoder.key
+                        for c in self.input_config.conditions]
+                if all(k in batch for k in keys):
+                    B = len(batch[keys[0]])
+                    self.rngs, key = self.rngs.get_random_key()
+                    mask = key.bernoulli((B,), self.unconditional_prob,
+                                         device="cpu").bool()
+                    conds = self.input_config.process_conditioning(batch, mask)
                     batch = dict(batch)
-                    batch["text_emb"] = cond.encoder.encode_from_tokens(batch[key]) \
-                        if cond.pretokenized else cond.encoder(batch[key])
+                    batch["cond_embs"] = tuple(conds)
         # video batches [B,T,H,W,C] fold time into batch for the 2D model path
         img = batch["image"]
         if torch.is_tensor(img) and img.dim() == 5 and not hasattr(self.model, "is_video_model"):

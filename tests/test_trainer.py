@@ -245,3 +245,54 @@ def test_validation_sample_with_cfg(tmp_path):
                                guidance_scale=3.0, conditioning_context=ctx)
     assert out.shape == (2, 16, 16, 3)
     assert torch.isfinite(out).all()
+
+
+def test_general_trainer_multi_condition_step(tmp_path):
+    """N-modality conditioning: two conditions, each
+    CFG-dropped against its OWN null embedding, reach the model as separate
+    context tensors (reference inputs/__init__.py:123-146)."""
+    from flaxdiff_amd.inputs import ConditionalInputConfig, DiffusionInputConfig
+    from flaxdiff_amd.inputs.encoders import DummyTextEncoder
+    from flaxdiff_amd.predictors import EpsilonPredictionTransform
+    from flaxdiff_amd.schedulers import CosineNoiseScheduler
+    from flaxdiff_amd.trainer import GeneralDiffusionTrainer
+
+    class TwoCondModel(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.proj1 = torch.nn.Linear(8, 4)
+            self.proj2 = torch.nn.Linear(6, 4)
+            self.out = torch.nn.Conv2d(3, 3, 1)
+            self.seen = []
+
+        def forward(self, x, temb, c_text, c_style):
+            self.seen.append((tuple(c_text.shape), tuple(c_style.shape)))
+            bias = self.proj1(c_text.float().mean(1)) + \
+                self.proj2(c_style.float().mean(1))
+            y = self.out(x.permute(0, 3, 1, 2)).permute(0, 2, 3, 1)
+            return y + bias.mean(-1)[:, None, None, None]
+
+    text_enc = DummyTextEncoder(seq_len=5, dim=8)
+    style_enc = DummyTextEncoder(seq_len=3, dim=6)
+    style_enc.key = "style"
+    cfg = DiffusionInputConfig(
+        sample_data_key="image", sample_data_shape=(16, 16, 3),
+        conditions=[
+            ConditionalInputConfig(encoder=text_enc, unconditional_input=""),
+            ConditionalInputConfig(encoder=style_enc,
+                                   conditioning_data_key="style",
+                                   unconditional_input=""),
+        ])
+    model = TwoCondModel()
+    tr = GeneralDiffusionTrainer(
+        model, CosineNoiseScheduler(1000), EpsilonPredictionTransform(),
+        input_config=cfg, name="multi-cond",
+        checkpoint_base_path=str(tmp_path), distributed=False)
+    batch = {
+        "image": torch.randint(0, 255, (4, 16, 16, 3), dtype=torch.uint8),
+        "text": ["a", "b", "c", "d"],
+        "style": ["s1", "s2", "s3", "s4"],
+    }
+    out = tr.train_step(batch)
+    assert "loss" in out and out["loss"] == out["loss"]
+    assert model.seen[-1] == ((4, 5, 8), (4, 3, 6))
