@@ -14,6 +14,7 @@ import torch
 
 from flaxdiff_amd.inputs import (ConditionalInputConfig, DiffusionInputConfig,
                                  DummyTextEncoder)
+from flaxdiff_amd.metrics import EvaluationMetric
 from flaxdiff_amd.models import Unet
 from flaxdiff_amd.predictors import KarrasPredictionTransform
 from flaxdiff_amd.samplers import EulerAncestralSampler
@@ -67,14 +68,28 @@ def main():
                                           generator=g, dtype=torch.uint8),
                    "text": toks}
 
-    trainer.train_loop(batches(), steps=args.steps)
-    trainer.save(config={"architecture": "unet", "model": model_cfg,
-                         "noise_schedule": "edm",
-                         "input_config": input_config.serialize(),
-                         "arguments": {"image_size": args.size}}, block=True)
+    # epoch-end validation: real multi-step EMA sampling + an eval metric
+    # with best-direction tracking (reference general_diffusion_trainer
+    # :378-519). pixel_std flags degenerate (collapsed/blank) samples.
+    trainer.eval_metrics = [EvaluationMetric(
+        function=lambda samples, batch: float(samples.float().std()),
+        name="pixel_std", higher_is_better=True)]
+    val_fn = trainer.make_validation_fn(
+        sampler_class=EulerAncestralSampler, num_samples=args.batch,
+        resolution=args.size, diffusion_steps=10, guidance_scale=3.0,
+        conditioning_context=encoder(captions[: args.batch]))
+    cfg = {"architecture": "unet", "model": model_cfg,
+           "noise_schedule": "edm",
+           "input_config": input_config.serialize(),
+           "arguments": {"image_size": args.size}}
+    trainer.fit(batches(), steps_per_epoch=args.steps, epochs=1,
+                val_fn=val_fn, config=cfg)
+    print("val metrics (best):", trainer.best_metric_values)
+    assert trainer.best_metric_values.get("pixel_std", 0) > 0
+
     out = trainer.validation_sample(
         EulerAncestralSampler, num_samples=args.batch, resolution=args.size,
-        diffusion_steps=5, guidance_scale=3.0,
+        diffusion_steps=10, guidance_scale=3.0,
         conditioning_context=encoder(captions[: args.batch]))
     print("CFG samples:", tuple(out.shape))
 

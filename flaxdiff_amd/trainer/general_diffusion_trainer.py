@@ -61,7 +61,9 @@ class GeneralDiffusionTrainer(DiffusionTrainer):
                 keys = [c.conditioning_data_key or c.encoder.key
                         for c in self.input_config.conditions]
                 if all(k in batch for k in keys):
-                    B = len(batch[keys[0]])
+                    # batch size from the SAMPLE tensor (a tokenizer
+                    # BatchEncoding's len() is its key count, not B)
+                    B = len(batch.get("image", batch[sample_key]))
                     self.rngs, key = self.rngs.get_random_key()
                     mask = key.bernoulli((B,), self.unconditional_prob,
                                          device="cpu").bool()
