@@ -25,38 +25,59 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
+from .. import ops
 from .attention import TransformerBlock
 from .common import (Conv, Dense, Downsample, FourierEmbedding, GroupNorm,
                      ResidualBlock, TimeProjection, Upsample)
 
 
-class TemporalConvLayer(nn.Module):
-    """4x [GN -> SiLU -> (3,1,1) conv] with zero-init final, residual
-    (reference unet_3d_blocks.py:103-167)."""
+class _TemporalConv(nn.Module):
+    """One FULL (3,1,1) conv over frames: three [Cin,Cout] taps applied to
+    the previous/current/next frame (zero-padded edges) — exactly a 1-D
+    temporal conv with complete channel mixing, expressed as three dense
+    MFMA GEMMs over the folded pixels (reference uses nn.Conv k=(3,1,1))."""
 
-    def __init__(self, channels: int, norm_groups: int = 32):
+    def __init__(self, cin: int, cout: int, zero_init: bool = False):
         super().__init__()
-        g = min(norm_groups, channels)
-        while channels % g:
-            g -= 1
-        self.norms = nn.ModuleList([GroupNorm(g, channels, eps=1e-5)
-                                    for _ in range(4)])
-        self.convs = nn.ParameterList()
-        self.biases = nn.ParameterList()
-        for i in range(4):
-            w = torch.zeros(3, channels) if i == 3 else \
-                torch.randn(3, channels) * (1.0 / (3 * channels) ** 0.5)
-            self.convs.append(nn.Parameter(w))
-            self.biases.append(nn.Parameter(torch.zeros(channels)))
+        def mk():
+            if zero_init:
+                return nn.Parameter(torch.zeros(cin, cout))
+            w = torch.randn(cin, cout) * (1.0 / (3 * cin)) ** 0.5
+            return nn.Parameter(w)
+        self.w_prev = mk()
+        self.w_cur = mk()
+        self.w_next = mk()
+        self.bias = nn.Parameter(torch.zeros(cout))
 
-    def _tconv(self, x: torch.Tensor, w: torch.Tensor, b: torch.Tensor,
-               T: int) -> torch.Tensor:
-        # x: [B, T, H, W, C]; depthwise temporal kernel w: [3, C]
-        w = w.to(x.dtype)
-        # SAME zero padding (frame edges contribute zero, like the reference)
-        prev = torch.cat([torch.zeros_like(x[:, :1]), x[:, :-1]], dim=1)
-        nxt = torch.cat([x[:, 1:], torch.zeros_like(x[:, :1])], dim=1)
-        return prev * w[0] + x * w[1] + nxt * w[2] + b.to(x.dtype)
+    def forward(self, v: torch.Tensor) -> torch.Tensor:
+        # v: [B, T, H, W, Cin]
+        prev = torch.cat([torch.zeros_like(v[:, :1]), v[:, :-1]], dim=1)
+        nxt = torch.cat([v[:, 1:], torch.zeros_like(v[:, :1])], dim=1)
+        out = ops.dense(prev, self.w_prev) + ops.dense(v, self.w_cur) \
+            + ops.dense(nxt, self.w_next)
+        return out + self.bias.to(out.dtype)
+
+
+class TemporalConvLayer(nn.Module):
+    """4x [GN -> SiLU -> full (3,1,1) conv] with ZERO-INIT final conv so the
+    block starts as identity, residual (reference unet_3d_blocks.py:103-167:
+    conv1 in->out, conv2 out->in, conv3 in->in, conv4 in->in zero-init)."""
+
+    def __init__(self, channels: int, norm_groups: int = 32,
+                 out_channels: Optional[int] = None):
+        super().__init__()
+        cout = out_channels or channels
+        chans = [(channels, cout), (cout, channels),
+                 (channels, channels), (channels, channels)]
+        def gn(c):
+            g = min(norm_groups, c)
+            while c % g:
+                g -= 1
+            return GroupNorm(g, c, eps=1e-5)
+        self.norms = nn.ModuleList([gn(ci) for ci, _ in chans])
+        self.convs = nn.ModuleList(
+            [_TemporalConv(ci, co, zero_init=(i == 3))
+             for i, (ci, co) in enumerate(chans)])
 
     def forward(self, x: torch.Tensor, num_frames: int) -> torch.Tensor:
         # x: [B*T, H, W, C]
@@ -64,10 +85,9 @@ class TemporalConvLayer(nn.Module):
         B = BT // num_frames
         v = x.reshape(B, num_frames, H, W, C)
         identity = v
-        for norm, w, b in zip(self.norms, self.convs, self.biases):
-            h = norm(v.reshape(BT, H, W, C), silu=True).reshape(B, num_frames,
-                                                                H, W, C)
-            v = self._tconv(h, w, b, num_frames)
+        for norm, conv in zip(self.norms, self.convs):
+            h = norm(v.reshape(-1, H, W, v.shape[-1]), silu=True)
+            v = conv(h.reshape(B, num_frames, H, W, -1))
         return (identity + v).reshape(BT, H, W, C)
 
 

@@ -22,8 +22,9 @@ def test_temporal_attention_identity_at_init():
 
 def test_temporal_conv_mixes_frames():
     tc = TemporalConvLayer(8, norm_groups=4)
-    with torch.no_grad():
-        tc.convs[3].fill_(0.1)  # activate the final conv
+    with torch.no_grad():  # activate the (zero-init) final conv
+        for w in (tc.convs[3].w_prev, tc.convs[3].w_cur, tc.convs[3].w_next):
+            w.fill_(0.1)
     x = torch.randn(1 * 4, 2, 2, 8)
     y = tc(x, num_frames=4)
     x2 = x.clone().reshape(1, 4, 2, 2, 8)
@@ -68,3 +69,62 @@ def test_general_trainer_keeps_video_shape():
                                     dtype=torch.uint8)}
     out = tr.train_step(batch)
     assert out["loss"] == out["loss"]
+
+
+def test_temporal_conv_identity_at_init_and_channel_mixing():
+    """Reference parity (unet_3d_blocks.py:103-167): the zero-init final
+    (3,1,1) conv makes the layer exact identity at init, and the convs are
+    FULL channel-mixing (not depthwise)."""
+    torch.manual_seed(0)
+    tc = TemporalConvLayer(8, norm_groups=4)
+    x = torch.randn(2 * 4, 2, 2, 8)
+    assert torch.equal(tc(x, num_frames=4), x)     # identity at init
+
+    with torch.no_grad():
+        tc.convs[3].w_cur.zero_()
+        tc.convs[3].w_cur[0, 3] = 1.0              # channel 0 -> channel 3
+    y0 = tc(x, num_frames=4)
+    x2 = x.clone()
+    x2[..., 0] += 3.0                              # perturb channel 0 only
+    y1 = tc(x2, num_frames=4)
+    d = (y1 - y0).abs()
+    assert d[..., 3].max() > 0                     # leaked into channel 3
+
+
+def test_unet3d_reference_block_structure_and_init_independence():
+    """Structural parity with the reference's CrossAttn 3-D blocks
+    (unet_3d_blocks.py:170-505): every down/up layer runs
+    resnet -> temporal conv -> spatial cross-attention -> temporal attention,
+    and every temporal mixer is zero-init — so at init the video model is
+    exactly a per-frame 2-D model (frame t's output ignores other frames)."""
+    import torch.nn as nn
+    from flaxdiff_amd.models.common import ResidualBlock
+    from flaxdiff_amd.models.attention import TransformerBlock
+    from flaxdiff_amd.models.unet_3d import TemporalAttention
+
+    torch.manual_seed(0)
+    model = UNet3D(emb_features=32, feature_depths=(8, 16),
+                   attention_configs=({"heads": 2}, {"heads": 2}),
+                   num_res_blocks=1, norm_groups=4, context_dim=16)
+    for level in list(model.down) + list(model.up):
+        for rb, tc, ab, tb in zip(level["res"], level["tconv"],
+                                  level["attn"], level["tattn"]):
+            assert isinstance(rb, ResidualBlock)
+            assert isinstance(tc, TemporalConvLayer)
+            assert isinstance(ab, (TransformerBlock, nn.Identity))
+            assert isinstance(tb, (TemporalAttention, nn.Identity))
+    assert isinstance(model.mid_tconv, TemporalConvLayer)
+    assert isinstance(model.mid_attn, TransformerBlock)
+
+    x = torch.randn(1, 4, 16, 16, 3)
+    t = torch.rand(1)
+    ctx = torch.randn(1, 5, 16)
+    with torch.no_grad():
+        y0 = model(x, t, ctx)
+        x2 = x.clone()
+        x2[:, 2] += 5.0
+        y1 = model(x2, t, ctx)
+    d = (y1 - y0).abs().amax(dim=(0, 2, 3, 4))
+    assert d[2] > 0
+    assert d[0] == 0 and d[1] == 0 and d[3] == 0, \
+        "temporal mixers not identity at init (zero-init parity broken)"
