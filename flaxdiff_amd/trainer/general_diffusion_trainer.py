@@ -123,9 +123,20 @@ class GeneralDiffusionTrainer(DiffusionTrainer):
         return val_fn
 
     # ------------------------------------------------------------------
-    def push_to_registry(self, registry_dir: str = "./registry"):
-        """Export the best checkpoint locally (stand-in for the reference's
-        wandb model-registry push, :560-594)."""
+    def push_to_registry(self, registry_dir: str = "./registry",
+                         compare_metric: Optional[str] = None,
+                         aliases: Optional[List[str]] = None):
+        """Export the latest checkpoint to a local model registry (stand-in
+        for the reference's wandb registry push + best-run comparison,
+        general_diffusion_trainer.py:560-703).
+
+        With compare_metric set, the push only replaces the registry entry
+        when this run's best value of that metric beats the stored one
+        (direction from eval_metrics); the registry keeps a meta.json with
+        the winning run's metrics, and "best"/alias symlink-style dirs.
+        Returns the destination path, or None when the incumbent wins.
+        """
+        import json
         import shutil
         from pathlib import Path
         if not self.dist.is_main:
@@ -134,10 +145,44 @@ class GeneralDiffusionTrainer(DiffusionTrainer):
         latest = self._ckpt.latest_step()
         if latest is None:
             return None
+
+        root = Path(registry_dir) / self.name
+        meta_path = root / "meta.json"
+        if compare_metric is not None and meta_path.exists():
+            prev = json.loads(meta_path.read_text())
+            prev_val = prev.get("metrics", {}).get(compare_metric)
+            cur_val = self.best_metric_values.get(compare_metric)
+            higher = True
+            for m in self.eval_metrics:
+                if m.name == compare_metric:
+                    higher = m.higher_is_better
+            if prev_val is not None and cur_val is not None:
+                if (cur_val <= prev_val) if higher else (cur_val >= prev_val):
+                    return None        # incumbent run stays the registry best
+
         src = self._ckpt.dir / str(latest)
-        dst = Path(registry_dir) / self.name / str(latest)
+        dst = root / str(latest)
         dst.parent.mkdir(parents=True, exist_ok=True)
         if dst.exists():
             shutil.rmtree(dst)
         shutil.copytree(src, dst)
+        meta = {"step": int(latest), "run": self.name,
+                "metrics": dict(self.best_metric_values),
+                "aliases": ["latest"] + list(aliases or [])}
+        meta_path.write_text(json.dumps(meta, indent=1))
         return str(dst)
+
+    @classmethod
+    def resume_from_registry(cls, registry_dir: str, name: str):
+        """Locate the registry's best checkpoint for `name` (the reference's
+        wandb artifact auto-download resume becomes a local path lookup).
+        Returns (checkpoint_dir, meta dict) for
+        DiffusionInferencePipeline.from_checkpoint or trainer.load()."""
+        import json
+        from pathlib import Path
+        root = Path(registry_dir) / name
+        meta_path = root / "meta.json"
+        if not meta_path.exists():
+            return None, None
+        meta = json.loads(meta_path.read_text())
+        return str(root / str(meta["step"])), meta
