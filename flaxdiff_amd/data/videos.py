@@ -209,3 +209,122 @@ def collate_video_batch(samples: List[Dict[str, Any]]) -> Dict[str, Any]:
     if "audio_mel" in samples[0]:
         batch["audio_mel"] = torch.stack([s["audio_mel"] for s in samples])
     return batch
+
+
+# ---------------------------------------------------------------------------
+# VoxCeleb2-style talking-face AV dataset (reference sources/voxceleb2.py:24-412)
+# ---------------------------------------------------------------------------
+
+def lower_half_mask(num_frames: int, height: int, width: int,
+                    pad: int = 0) -> np.ndarray:
+    """[F, H, W, 1] float mask: 1 on the (to-be-inpainted) lower face half,
+    shrunk by `pad` rows (reference voxceleb2.py:177-203 get_simple_mask)."""
+    m = np.zeros((num_frames, height, width, 1), dtype=np.float32)
+    m[:, max(height // 2 - pad, 0):, :, :] = 1.0
+    return m
+
+
+class VoxCeleb2LikeSource(DataSource):
+    """Talking-face AV clips in the voxceleb2 layout (id/.../clip.*), each
+    clip a video (any read_video_clip format) with optional audio:
+      * `<clip>.wav.npy` — raw waveform, or
+      * `.npz` clips with `frames` [T,H,W,3] uint8 and `audio` [S] float.
+
+    Per sample (reference voxceleb2.py:281-383):
+      instance_images [F,H,W,3]  — the training window
+      reference_images [F,H,W,3] — a DISJOINT window of the same identity
+      mask / instance_masks [F,H,W,1] — lower-half inpainting mask
+      instance_masked_images     — instance * (mask < 0.5)
+      mels [n_mels, T_a]         — log-mel of the window's audio (if present)
+      identity                   — speaker id (top-level dir name)
+    """
+
+    def __init__(self, root: str, num_frames: int = 8, resolution: int = 64,
+                 sample_rate: int = 16000, fps: float = 25.0,
+                 n_mels: int = 80, audio: bool = True):
+        self.root = Path(root)
+        self.num_frames = num_frames
+        self.resolution = resolution
+        self.sample_rate = sample_rate
+        self.fps = fps
+        self.n_mels = n_mels
+        self.audio = audio
+        exts = VideoFolderSource.VIDEO_EXTS + (".npz",)
+        self.items = sorted(p for p in self.root.rglob("*")
+                            if p.suffix.lower() in exts)
+        if not self.items:
+            raise FileNotFoundError(f"no clips under {root}")
+
+    def __len__(self):
+        return len(self.items)
+
+    def _load_clip(self, p: Path):
+        if p.suffix == ".npz":
+            z = np.load(p)
+            return np.asarray(z["frames"], dtype=np.uint8), \
+                (np.asarray(z["audio"], dtype=np.float32)
+                 if "audio" in z.files else None)
+        frames = read_video_clip(str(p))
+        wav = None
+        side = p.with_suffix(p.suffix + ".wav.npy")
+        alt = p.with_name(p.stem + ".wav.npy")
+        for cand in (side, alt):
+            if cand.exists():
+                wav = np.load(cand).astype(np.float32)
+                break
+        return frames, wav
+
+    def __getitem__(self, idx: int) -> Dict[str, Any]:
+        rng = np.random.default_rng(idx)
+        for attempt in range(10):
+            p = self.items[(idx + attempt) % len(self.items)]
+            frames, wav = self._load_clip(p)
+            F_ = self.num_frames
+            if frames.shape[0] >= 3 * F_:
+                break
+        else:
+            raise RuntimeError("no clip with >= 3x num_frames frames")
+
+        total = frames.shape[0]
+        start = int(rng.integers(F_ // 2, total - F_ - F_ // 2 + 1))
+        inst = frames[start:start + F_]
+        # disjoint reference window of the SAME identity (voxceleb2.py:205-242)
+        ref_lo = 0 if start >= F_ else start + F_
+        ref_hi = start - F_ if start >= F_ else total - F_
+        ref_start = int(rng.integers(ref_lo, max(ref_hi, ref_lo) + 1))
+        ref = frames[ref_start:ref_start + F_]
+
+        def _resize(clip):
+            t = torch.from_numpy(np.ascontiguousarray(clip)).permute(0, 3, 1, 2).float()
+            t = torch.nn.functional.interpolate(t, size=(self.resolution,
+                                                         self.resolution),
+                                                mode="bilinear",
+                                                align_corners=False)
+            return t.permute(0, 2, 3, 1).round().clamp(0, 255).byte().numpy()
+
+        inst = _resize(inst)
+        ref = _resize(ref)
+        mask = lower_half_mask(F_, self.resolution, self.resolution)
+        out: Dict[str, Any] = {
+            "instance_images": inst,
+            "reference_images": ref,
+            "mask": mask,
+            "instance_masks": mask,
+            "instance_masked_images": (inst.astype(np.float32)
+                                       * (mask < 0.5)).astype(np.uint8),
+            "identity": p.relative_to(self.root).parts[0]
+            if len(p.relative_to(self.root).parts) > 1 else p.stem,
+            "video": inst,      # generic video key for the existing augmenters
+            "caption": "",
+        }
+        if self.audio and wav is not None:
+            spf = self.sample_rate / self.fps      # samples per frame
+            a0 = int(start * spf)
+            a1 = int((start + F_) * spf)
+            window = wav[a0:min(a1, len(wav))]
+            if len(window) > 0:
+                out["raw_audio"] = window
+                out["mels"] = mel_spectrogram(
+                    torch.from_numpy(window), sample_rate=self.sample_rate,
+                    n_mels=self.n_mels).numpy()
+        return out

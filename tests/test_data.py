@@ -133,3 +133,78 @@ def test_trainer_consumes_loader_batches():
                           distributed=False)
     out = tr.train_step(next(iter(dl)))
     assert np.isfinite(out["loss"])
+
+
+def test_record_shards_roundtrip(tmp_path):
+    """ArrayRecord-equivalent local shard format: write -> random access ->
+    registry -> sharded DataLoader."""
+    import numpy as np
+    from flaxdiff_amd.data import (RecordShardDataset, RecordShardWriter,
+                                   RecordSource, get_dataset,
+                                   register_record_dataset,
+                                   write_records_from_source)
+    from flaxdiff_amd.data.sources import SyntheticImageSource
+
+    recs = []
+    for sh in range(2):
+        with RecordShardWriter(str(tmp_path / f"s{sh}.rec")) as w:
+            for i in range(5):
+                r = {"image": np.full((8, 8, 3), sh * 10 + i, dtype=np.uint8),
+                     "caption": f"cap-{sh}-{i}", "score": 0.5 + i}
+                recs.append(r)
+                w.write(r)
+
+    ds = RecordShardDataset(str(tmp_path))
+    assert len(ds) == 10
+    for i in (0, 4, 5, 9):
+        got = ds[i]
+        assert got["caption"] == recs[i]["caption"]
+        assert (got["image"] == recs[i]["image"]).all()
+        assert got["score"] == recs[i]["score"]
+
+    # registry + sharded loader (2 ranks see disjoint halves)
+    register_record_dataset("test-recs", str(tmp_path), image_size=8)
+    seen = set()
+    for rank in (0, 1):
+        dl = get_dataset("test-recs", global_batch_size=2, rank=rank,
+                         world_size=2, worker_count=0, shuffle=False)
+        for batch in dl:
+            # images are constant-filled: the fill value identifies the record
+            seen.update(int(v) for v in batch["image"][:, 0, 0, 0])
+    assert len(seen) == 10
+
+    # converter: any source -> shards
+    src = SyntheticImageSource(image_size=8, num_samples=7)
+    paths = write_records_from_source(src, str(tmp_path / "conv"),
+                                      shard_size=3)
+    assert len(paths) == 3
+    ds2 = RecordShardDataset(str(tmp_path / "conv"))
+    assert len(ds2) == 7
+    assert (ds2[6]["image"] == src[6]["image"]).all()
+
+
+def test_voxceleb2_like_av_source(tmp_path):
+    """VoxCeleb2-style AV samples: disjoint ref window, lower-half mask,
+    masked images, mel audio (reference sources/voxceleb2.py:281-383)."""
+    import numpy as np
+    from flaxdiff_amd.data import VoxCeleb2LikeSource
+
+    sr, fps, T = 16000, 25.0, 40
+    for ident in ("id001", "id002"):
+        d = tmp_path / ident
+        d.mkdir()
+        frames = np.random.randint(0, 255, (T, 32, 32, 3), dtype=np.uint8)
+        audio = np.random.randn(int(T / fps * sr)).astype(np.float32)
+        np.savez(d / "clip0.npz", frames=frames, audio=audio)
+
+    src = VoxCeleb2LikeSource(str(tmp_path), num_frames=8, resolution=16,
+                              sample_rate=sr, fps=fps, n_mels=20)
+    assert len(src) == 2
+    s = src[0]
+    assert s["instance_images"].shape == (8, 16, 16, 3)
+    assert s["reference_images"].shape == (8, 16, 16, 3)
+    assert s["mask"].shape == (8, 16, 16, 1)
+    assert s["mask"][:, :7].max() == 0 and s["mask"][:, 8:].min() == 1
+    assert (s["instance_masked_images"][:, 8:] == 0).all()
+    assert s["identity"] == "id001"
+    assert s["mels"].shape[0] == 20 and s["mels"].shape[1] > 10
