@@ -446,10 +446,11 @@ class _AttentionFn(torch.autograd.Function):
         # Small-KV fused MFMA flash backward: 3x the composed GEMM path at
         # the bench shapes (2.34 vs 7.12 ms at Sq=4096 D=16 — tools/
         # attn_bwd_ab.py). Larger head dims fall through to the GEMMs.
-        if k.shape[2] <= 128 and q.shape[3] <= 32:
+        use_v1 = os.environ.get("FD_ATTN_BWD_V1")
+        dmax = 32 if use_v1 else 64
+        if k.shape[2] <= 128 and q.shape[3] <= dmax:
             ext = _require_ext()
-            fn = ext.attn_bwd_smallkv if os.environ.get("FD_ATTN_BWD_V1") \
-                else ext.attn_bwd_smallkv_v2
+            fn = ext.attn_bwd_smallkv if use_v1 else ext.attn_bwd_smallkv_v2
             dq, dk, dv = fn(q, k, v, do.contiguous(), lse, scale)
             return dq, dk, dv, None
         # General shapes: recompute P row-exactly from the saved log-sum-exp,
