@@ -191,11 +191,12 @@ class Conv(nn.Module):
         self.weight = nn.Parameter(w)
         self.bias = nn.Parameter(torch.zeros(features)) if use_bias else None
 
-    def forward(self, x: torch.Tensor) -> torch.Tensor:
+    def forward(self, x: torch.Tensor, add: torch.Tensor = None) -> torch.Tensor:
         if x.is_cuda and x.dtype == torch.bfloat16:
-            return ops.conv2d(x, self.weight, self.bias, stride=self.stride)
+            return ops.conv2d(x, self.weight, self.bias, stride=self.stride,
+                              add=add)
         return ops.conv2d(x, _cast(self.weight, x.dtype), _cast(self.bias, x.dtype),
-                          stride=self.stride)
+                          stride=self.stride, add=add)
 
 
 class ConvTranspose(nn.Module):
@@ -362,11 +363,14 @@ class ResidualBlock(nn.Module):
         out = out + t[:, None, None, :]
 
         out = self._norm_act(self.norm2, out)
-        out = self.conv2(out)
-
         if self.residual_conv is not None:
             residual = self.residual_conv(residual)
-        out = out + residual
+        if isinstance(self.conv2, Conv) and residual.shape[-1] == self.features:
+            # residual add fused into the conv epilogue (saves one full
+            # HBM read+write elementwise pass per block on GPU)
+            out = self.conv2(out, add=residual)
+        else:
+            out = self.conv2(out) + residual
 
         if extra_features is not None:
             out = torch.cat([out, extra_features], dim=-1)

@@ -197,14 +197,18 @@ def _routable_param(p: Optional[torch.Tensor]) -> bool:
 
 class _Conv2dFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, w, b, stride):
+    def forward(ctx, x, w, b, stride, add):
         ext = _require_ext()
         wk = _kernel_view(w)
-        # fp32 bias feeds the kernel's fp32 epilogue directly (no cast)
-        y = ext.conv2d_fwd(x, wk, b if b is not None else torch.Tensor(), stride)
+        # fp32 bias feeds the kernel's fp32 epilogue directly (no cast);
+        # `add` (residual, same shape as y) is fused into the same epilogue
+        y = ext.conv2d_fwd(x, wk, b if b is not None else torch.Tensor(),
+                           stride,
+                           add if add is not None else torch.Tensor())
         ctx.save_for_backward(x, wk)
         ctx.stride = stride
         ctx.has_bias = b is not None
+        ctx.has_add = add is not None
         ctx.w_dtype = w.dtype
         ctx.b_dtype = b.dtype if b is not None else None
         return y
@@ -225,7 +229,8 @@ class _Conv2dFn(torch.autograd.Function):
             db = _bias_grad(dy)          # fp32
             if ctx.b_dtype != torch.float32:
                 db = db.to(ctx.b_dtype)
-        return dx, dw, db, None
+        dadd = dy if ctx.has_add else None   # y = conv + add -> d(add) = dy
+        return dx, dw, db, None, dadd
 
 
 class _DenseFn(torch.autograd.Function):
@@ -312,9 +317,12 @@ def _bias_grad(dy: torch.Tensor) -> torch.Tensor:
 
 
 def conv2d(x: torch.Tensor, w: torch.Tensor, b: Optional[torch.Tensor] = None,
-           stride: int = 1) -> torch.Tensor:
+           stride: int = 1, add: Optional[torch.Tensor] = None) -> torch.Tensor:
     """SAME-padded NHWC conv. 1x1 convs route to a plain GEMM.
-    w/b may be fp32 master params with bf16 shadows (no cast kernels)."""
+    w/b may be fp32 master params with bf16 shadows (no cast kernels).
+    `add` fuses a same-shape residual into the conv epilogue (ResidualBlock's
+    `conv2(out) + residual`, reference common.py:268) — one fewer full
+    HBM read+write pass per block on GPU; falls back to y + add elsewhere."""
     kh, kw = w.shape[0], w.shape[1]
     if kh == 1 and kw == 1 and stride == 1:
         w2 = w.reshape(w.shape[2], w.shape[3])
@@ -328,15 +336,24 @@ def conv2d(x: torch.Tensor, w: torch.Tensor, b: Optional[torch.Tensor] = None,
             if sht is not None:
                 w2._shadow_bf16_t = sht
         y = dense(x.reshape(-1, w.shape[2]), w2, b)
-        return y.reshape(*x.shape[:-1], w.shape[3])
+        y = y.reshape(*x.shape[:-1], w.shape[3])
+        return y if add is None else y + add
     if _use_hip(x) and x.dtype == torch.bfloat16 and _routable_param(w):
+        if add is not None and (add.dtype != torch.bfloat16 or stride != 1):
+            out = _Conv2dFn.apply(x.contiguous(), w.contiguous(),
+                                  b.contiguous() if b is not None else None,
+                                  stride, None)
+            return out + add
         return _Conv2dFn.apply(x.contiguous(), w.contiguous(),
-                               b.contiguous() if b is not None else None, stride)
+                               b.contiguous() if b is not None else None,
+                               stride,
+                               add.contiguous() if add is not None else None)
     if w.dtype != x.dtype:
         w = w.to(x.dtype)
     if b is not None and b.dtype != x.dtype:
         b = b.to(x.dtype)
-    return reference.conv2d_nhwc(x, w, b, stride=stride, padding="same")
+    y = reference.conv2d_nhwc(x, w, b, stride=stride, padding="same")
+    return y if add is None else y + add
 
 
 class _ConvTransposeFn(torch.autograd.Function):
@@ -365,7 +382,8 @@ class _ConvTransposeFn(torch.autograd.Function):
         x, wf = ctx.saved_tensors
         ext = _require_ext()
         dout = dout.contiguous()
-        dx = ext.conv2d_fwd(dout, wf, torch.Tensor(), ctx.stride) \
+        dx = ext.conv2d_fwd(dout, wf, torch.Tensor(), ctx.stride,
+                            torch.Tensor()) \
             if ctx.needs_input_grad[0] else None
         dwf = None
         if ctx.needs_input_grad[1]:

@@ -159,6 +159,40 @@ def test_conv2d_fwd(B, H, W, Ci, Co, k, st):
     assert err < 3e-2, f"conv fwd rel err {err}"
 
 
+@pytest.mark.parametrize("B,H,W,Ci,Co", [
+    (2, 16, 16, 64, 128),        # general igemm path
+    (2, 8, 8, 512, 512),         # halo path (v2-eligible)
+    (2, 64, 64, 64, 64),         # level-0 halo v1 path
+    (2, 48, 48, 32, 32),         # general kernel, Co not mult of 4 tails
+])
+def test_conv2d_fused_residual_add(B, H, W, Ci, Co):
+    """conv2d(add=residual) == conv2d() + residual, fwd and the dadd=dy
+    pass-through grad (ResidualBlock epilogue fusion)."""
+    torch.manual_seed(11)
+    x = (torch.randn(B, H, W, Ci) * 0.5).bfloat16().to(_dev())
+    w = (torch.randn(3, 3, Ci, Co) / (9 * Ci) ** 0.5).bfloat16().to(_dev())
+    bias = (torch.randn(Co) * 0.1).bfloat16().to(_dev())
+    res = torch.randn(B, H, W, Co).bfloat16().to(_dev())
+
+    y0 = ops.conv2d(x, w, bias, stride=1) + res
+    y1 = ops.conv2d(x, w, bias, stride=1, add=res)
+    assert rel_err(y1.float().cpu(), y0.float().cpu()) < 1e-3
+
+    xa = x.clone().requires_grad_(True)
+    wa = w.clone().requires_grad_(True)
+    ra = res.clone().requires_grad_(True)
+    ya = ops.conv2d(xa, wa, bias, stride=1, add=ra)
+    dy = torch.randn_like(ya)
+    ya.backward(dy)
+    assert torch.equal(ra.grad, dy)          # dadd is exactly dy
+    xb = x.clone().requires_grad_(True)
+    wb = w.clone().requires_grad_(True)
+    yb = ops.conv2d(xb, wb, bias, stride=1) + res
+    yb.backward(dy)
+    assert rel_err(xa.grad.float().cpu(), xb.grad.float().cpu()) < 1e-5
+    assert rel_err(wa.grad.float().cpu(), wb.grad.float().cpu()) < 1e-5
+
+
 @pytest.mark.parametrize("B,H,W,Ci,Co,k,st", CONV_SHAPES[:6] + CONV_SHAPES[8:12])
 def test_conv2d_backward(B, H, W, Ci, Co, k, st):
     torch.manual_seed(4)
