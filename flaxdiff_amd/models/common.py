@@ -191,12 +191,13 @@ class Conv(nn.Module):
         self.weight = nn.Parameter(w)
         self.bias = nn.Parameter(torch.zeros(features)) if use_bias else None
 
-    def forward(self, x: torch.Tensor, add: torch.Tensor = None) -> torch.Tensor:
+    def forward(self, x: torch.Tensor, add: torch.Tensor = None,
+                badd: torch.Tensor = None) -> torch.Tensor:
         if x.is_cuda and x.dtype == torch.bfloat16:
             return ops.conv2d(x, self.weight, self.bias, stride=self.stride,
-                              add=add)
+                              add=add, badd=badd)
         return ops.conv2d(x, _cast(self.weight, x.dtype), _cast(self.bias, x.dtype),
-                          stride=self.stride, add=add)
+                          stride=self.stride, add=add, badd=badd)
 
 
 class ConvTranspose(nn.Module):
@@ -357,10 +358,13 @@ class ResidualBlock(nn.Module):
                 textemb: torch.Tensor = None, extra_features: torch.Tensor = None):
         residual = x
         out = self._norm_act(self.norm1, x)
-        out = self.conv1(out)
-
         t = self.temb_projection(temb.to(out.dtype))
-        out = out + t[:, None, None, :]
+        if isinstance(self.conv1, Conv):
+            # temb broadcast fused into the conv epilogue (one fewer full
+            # HBM read+write elementwise pass per block)
+            out = self.conv1(out, badd=t)
+        else:
+            out = self.conv1(out) + t[:, None, None, :]
 
         out = self._norm_act(self.norm2, out)
         if self.residual_conv is not None:

@@ -197,18 +197,21 @@ def _routable_param(p: Optional[torch.Tensor]) -> bool:
 
 class _Conv2dFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, w, b, stride, add):
+    def forward(ctx, x, w, b, stride, add, badd):
         ext = _require_ext()
         wk = _kernel_view(w)
         # fp32 bias feeds the kernel's fp32 epilogue directly (no cast);
-        # `add` (residual, same shape as y) is fused into the same epilogue
+        # `add` (residual, same shape as y) and `badd` (per-sample [B,C]
+        # broadcast, the temb projection) are fused into the same epilogue
         y = ext.conv2d_fwd(x, wk, b if b is not None else torch.Tensor(),
                            stride,
-                           add if add is not None else torch.Tensor())
+                           add if add is not None else torch.Tensor(),
+                           badd if badd is not None else torch.Tensor())
         ctx.save_for_backward(x, wk)
         ctx.stride = stride
         ctx.has_bias = b is not None
         ctx.has_add = add is not None
+        ctx.has_badd = badd is not None
         ctx.w_dtype = w.dtype
         ctx.b_dtype = b.dtype if b is not None else None
         return y
@@ -230,7 +233,11 @@ class _Conv2dFn(torch.autograd.Function):
             if ctx.b_dtype != torch.float32:
                 db = db.to(ctx.b_dtype)
         dadd = dy if ctx.has_add else None   # y = conv + add -> d(add) = dy
-        return dx, dw, db, None, dadd
+        dbadd = None
+        if ctx.has_badd:                     # d(badd)[b,c] = sum_hw dy
+            B = dy.shape[0]
+            dbadd = dy.reshape(B, -1, dy.shape[-1]).sum(1)
+        return dx, dw, db, None, dadd, dbadd
 
 
 class _DenseFn(torch.autograd.Function):
@@ -317,12 +324,15 @@ def _bias_grad(dy: torch.Tensor) -> torch.Tensor:
 
 
 def conv2d(x: torch.Tensor, w: torch.Tensor, b: Optional[torch.Tensor] = None,
-           stride: int = 1, add: Optional[torch.Tensor] = None) -> torch.Tensor:
+           stride: int = 1, add: Optional[torch.Tensor] = None,
+           badd: Optional[torch.Tensor] = None) -> torch.Tensor:
     """SAME-padded NHWC conv. 1x1 convs route to a plain GEMM.
     w/b may be fp32 master params with bf16 shadows (no cast kernels).
     `add` fuses a same-shape residual into the conv epilogue (ResidualBlock's
-    `conv2(out) + residual`, reference common.py:268) — one fewer full
-    HBM read+write pass per block on GPU; falls back to y + add elsewhere."""
+    `conv2(out) + residual`, reference common.py:268); `badd` [B, Cout] fuses
+    the per-sample temb-projection broadcast (`out + t[:,None,None,:]`,
+    reference common.py:257) — each saves one full HBM read+write pass per
+    block on GPU; both fall back to plain adds elsewhere."""
     kh, kw = w.shape[0], w.shape[1]
     if kh == 1 and kw == 1 and stride == 1:
         w2 = w.reshape(w.shape[2], w.shape[3])
@@ -337,22 +347,35 @@ def conv2d(x: torch.Tensor, w: torch.Tensor, b: Optional[torch.Tensor] = None,
                 w2._shadow_bf16_t = sht
         y = dense(x.reshape(-1, w.shape[2]), w2, b)
         y = y.reshape(*x.shape[:-1], w.shape[3])
+        if badd is not None:
+            y = y + badd[:, None, None, :].to(y.dtype)
         return y if add is None else y + add
     if _use_hip(x) and x.dtype == torch.bfloat16 and _routable_param(w):
-        if add is not None and (add.dtype != torch.bfloat16 or stride != 1):
-            out = _Conv2dFn.apply(x.contiguous(), w.contiguous(),
-                                  b.contiguous() if b is not None else None,
-                                  stride, None)
-            return out + add
-        return _Conv2dFn.apply(x.contiguous(), w.contiguous(),
-                               b.contiguous() if b is not None else None,
-                               stride,
-                               add.contiguous() if add is not None else None)
+        if add is not None and add.dtype != torch.bfloat16:
+            add_in, add_out = None, add
+        else:
+            add_in, add_out = add, None
+        if badd is not None and badd.dtype != torch.bfloat16:
+            badd_in, badd_out = None, badd
+        else:
+            badd_in, badd_out = badd, None
+        out = _Conv2dFn.apply(
+            x.contiguous(), w.contiguous(),
+            b.contiguous() if b is not None else None, stride,
+            add_in.contiguous() if add_in is not None else None,
+            badd_in.contiguous() if badd_in is not None else None)
+        if badd_out is not None:
+            out = out + badd_out[:, None, None, :].to(out.dtype)
+        if add_out is not None:
+            out = out + add_out
+        return out
     if w.dtype != x.dtype:
         w = w.to(x.dtype)
     if b is not None and b.dtype != x.dtype:
         b = b.to(x.dtype)
     y = reference.conv2d_nhwc(x, w, b, stride=stride, padding="same")
+    if badd is not None:
+        y = y + badd[:, None, None, :].to(y.dtype)
     return y if add is None else y + add
 
 
@@ -383,7 +406,7 @@ class _ConvTransposeFn(torch.autograd.Function):
         ext = _require_ext()
         dout = dout.contiguous()
         dx = ext.conv2d_fwd(dout, wf, torch.Tensor(), ctx.stride,
-                            torch.Tensor()) \
+                            torch.Tensor(), torch.Tensor()) \
             if ctx.needs_input_grad[0] else None
         dwf = None
         if ctx.needs_input_grad[1]:
