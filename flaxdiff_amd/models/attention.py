@@ -38,7 +38,10 @@ class NormalAttention(nn.Module):
         self.to_v = Dense(cdim, inner, use_bias=use_bias)
         self.to_out = Dense(inner, query_dim, use_bias=use_bias)
 
-    def forward(self, x: torch.Tensor, context: Optional[torch.Tensor] = None):
+    def forward(self, x: torch.Tensor, context: Optional[torch.Tensor] = None,
+                add: Optional[torch.Tensor] = None):
+        # `add` (same shape as x) rides the out-projection's GEMM epilogue —
+        # the pre-norm residual add costs no extra kernel.
         orig_shape = x.shape
         if x.dim() == 4:
             B, H, W, C = x.shape
@@ -55,7 +58,7 @@ class NormalAttention(nn.Module):
 
         o = ops.attention(q, k, v)
         o = o.permute(0, 2, 1, 3).reshape(B, S, self.heads * self.dim_head)
-        o = self.to_out(o)
+        o = self.to_out(o, add=add.reshape(B, S, -1) if add is not None else None)
         return o.reshape(orig_shape)
 
 
@@ -85,8 +88,8 @@ class FeedForward(nn.Module):
         self.net_0 = GEGLU(dim)
         self.net_2 = Dense(dim * 4, dim)
 
-    def forward(self, x):
-        return self.net_2(self.net_0(x))
+    def forward(self, x, add=None):
+        return self.net_2(self.net_0(x), add=add)
 
 
 class BasicTransformerBlock(nn.Module):
@@ -116,12 +119,15 @@ class BasicTransformerBlock(nn.Module):
             self.norm3 = RMSNorm(query_dim, eps=norm_epsilon)
 
     def forward(self, hidden_states, context=None):
+        # residual adds fused into each sub-block's final GEMM epilogue
         if self.only_pure_attention:
             return self.attention2(hidden_states, context)
         if not self.use_cross_only:
-            hidden_states = hidden_states + self.attention1(self.norm1(hidden_states))
-        hidden_states = hidden_states + self.attention2(self.norm2(hidden_states), context)
-        hidden_states = hidden_states + self.ff(self.norm3(hidden_states))
+            hidden_states = self.attention1(self.norm1(hidden_states),
+                                            add=hidden_states)
+        hidden_states = self.attention2(self.norm2(hidden_states), context,
+                                        add=hidden_states)
+        hidden_states = self.ff(self.norm3(hidden_states), add=hidden_states)
         return hidden_states
 
 
@@ -162,7 +168,9 @@ class TransformerBlock(nn.Module):
             context = projected
         projected = self.block(projected, context)
         if self.use_projection:
-            projected = self.project_out(projected)
+            if self.only_pure_attention or self.explicitly_add_residual:
+                return self.project_out(projected, add=x)
+            return self.project_out(projected)
         if self.only_pure_attention or self.explicitly_add_residual:
             projected = x + projected
         return projected

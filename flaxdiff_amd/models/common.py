@@ -90,13 +90,13 @@ class Dense(nn.Module):
         self.weight = nn.Parameter(w)
         self.bias = nn.Parameter(torch.zeros(out_features)) if use_bias else None
 
-    def forward(self, x: torch.Tensor) -> torch.Tensor:
+    def forward(self, x: torch.Tensor, add: torch.Tensor = None) -> torch.Tensor:
         if x.is_cuda and x.dtype == torch.bfloat16:
             # fp32 masters go straight in: ops.dense uses the bf16 shadow for
             # compute and returns fp32 grads (no per-call cast kernels)
-            return ops.dense(x, self.weight, self.bias)
+            return ops.dense(x, self.weight, self.bias, add=add)
         return ops.dense(x, _cast(self.weight, x.dtype),
-                         _cast(self.bias, x.dtype))
+                         _cast(self.bias, x.dtype), add=add)
 
 
 # ---------------------------------------------------------------------------
@@ -115,6 +115,12 @@ class GroupNorm(nn.Module):
         self.bias = nn.Parameter(torch.zeros(num_channels))
 
     def forward(self, x: torch.Tensor, silu: bool = False) -> torch.Tensor:
+        if x.is_cuda:
+            # fp32 masters go straight to the HIP kernel (it consumes fp32
+            # gamma/beta and emits fp32 dgamma/dbeta) — no bf16 shadow round
+            # trip, no per-call cast kernels
+            return ops.group_norm(x, self.num_groups, self.weight, self.bias,
+                                  self.eps, silu)
         return ops.group_norm(x, self.num_groups, _cast(self.weight, x.dtype),
                               _cast(self.bias, x.dtype), self.eps, silu)
 
@@ -126,6 +132,8 @@ class RMSNorm(nn.Module):
         self.weight = nn.Parameter(torch.ones(dim))
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if x.is_cuda:
+            return ops.rms_norm(x, self.weight, self.eps)
         return ops.rms_norm(x, _cast(self.weight, x.dtype), self.eps)
 
 

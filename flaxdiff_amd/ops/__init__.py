@@ -248,23 +248,27 @@ class _DenseFn(torch.autograd.Function):
     through the split-M kernel (profiles/ MT64x64x256 rows)."""
 
     @staticmethod
-    def forward(ctx, x2d, w2d, b):
+    def forward(ctx, x2d, w2d, b, add):
         wk = _kernel_view(w2d)
         if not _GEMM_LIB and x2d.shape[1] % 64 == 0:
-            # hand-written MFMA GEMM (fused fp32 bias) — no hipBLASLt on the
-            # declared hot path (BASELINE.json north star). The weight comes
-            # in transposed ([N,K] NT layout, the fast A-row read).
+            # hand-written MFMA GEMM (fused fp32 bias + optional residual) —
+            # no hipBLASLt on the declared hot path (BASELINE.json north
+            # star). The weight comes in transposed ([N,K] NT layout).
             ext = _require_ext()
             bf = b if b is not None else torch.Tensor()
             if bf.numel() and bf.dtype != torch.float32:
                 bf = bf.float()
-            y = ext.gemm_nt(x2d, _kernel_view_t(w2d), bf)
+            y = ext.gemm_nt(x2d, _kernel_view_t(w2d), bf,
+                            add if add is not None else torch.Tensor())
         else:
             y = torch.matmul(x2d, wk)
             if b is not None:
                 y = y + _kernel_view(b)
+            if add is not None:
+                y = y + add
         ctx.save_for_backward(x2d, wk)
         ctx.has_bias = b is not None
+        ctx.has_add = add is not None
         ctx.w_dtype = w2d.dtype
         ctx.b_dtype = b.dtype if b is not None else None
         return y
@@ -294,18 +298,28 @@ class _DenseFn(torch.autograd.Function):
             db = _bias_grad(dy)
             if ctx.b_dtype != torch.float32:
                 db = db.to(ctx.b_dtype)
-        return dx, dw, db
+        dadd = dy if ctx.has_add else None   # y = xW + add -> d(add) = dy
+        return dx, dw, db, dadd
 
 
-def dense(x: torch.Tensor, w: torch.Tensor, b: Optional[torch.Tensor] = None
-          ) -> torch.Tensor:
-    """x [..., Cin] @ w [Cin, Cout] (+ b): library GEMM forward, split-M
-    wgrad backward on GPU. w/b may be fp32 masters with bf16 shadows."""
+def dense(x: torch.Tensor, w: torch.Tensor, b: Optional[torch.Tensor] = None,
+          add: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """x [..., Cin] @ w [Cin, Cout] (+ b): MFMA GEMM forward, split-M
+    wgrad backward on GPU. w/b may be fp32 masters with bf16 shadows.
+    `add` fuses a same-shape residual into the GEMM epilogue (transformer
+    residual adds, reference attention.py:240-303) in fp32 before the single
+    bf16 rounding."""
     if _use_hip(x) and x.dtype == torch.bfloat16 and _routable_param(w) \
             and _routable_param(b):
         lead = x.shape[:-1]
-        y = _DenseFn.apply(x.reshape(-1, x.shape[-1]).contiguous(), w, b)
-        return y.reshape(*lead, w.shape[1])
+        add2 = None
+        if add is not None and add.dtype == torch.bfloat16:
+            add2 = add.reshape(-1, w.shape[1]).contiguous()
+        y = _DenseFn.apply(x.reshape(-1, x.shape[-1]).contiguous(), w, b, add2)
+        y = y.reshape(*lead, w.shape[1])
+        if add is not None and add2 is None:
+            y = y + add
+        return y
     if w.dtype != x.dtype:
         w = w.to(x.dtype)
     y = torch.matmul(x, w)

@@ -234,6 +234,48 @@ def test_conv2d_fused_broadcast_add(B, H, W, Ci, Co, st):
     assert rel_err(xa.grad.float().cpu(), xb.grad.float().cpu()) < 1e-5
 
 
+def test_dense_fused_residual_add():
+    """dense(add=residual) == dense() + residual (transformer residual
+    fusion in the gemm_nt epilogue), with dadd = dy pass-through."""
+    torch.manual_seed(13)
+    M, K, N = 512, 128, 128
+    x = (torch.randn(M, K) * 0.5).bfloat16().to(_dev())
+    w = (torch.randn(K, N) / K ** 0.5).bfloat16().to(_dev())
+    b = (torch.randn(N) * 0.1).bfloat16().to(_dev())
+    res = torch.randn(M, N).bfloat16().to(_dev())
+
+    ref = x.float().cpu() @ w.float().cpu() + b.float().cpu() + res.float().cpu()
+    y1 = ops.dense(x, w, b, add=res)
+    assert rel_err(y1.float().cpu(), ref) < 3e-2
+
+    xa = x.clone().requires_grad_(True)
+    ra = res.clone().requires_grad_(True)
+    ya = ops.dense(xa, w, b, add=ra)
+    dy = torch.randn_like(ya)
+    ya.backward(dy)
+    assert torch.equal(ra.grad, dy)
+    xb = x.clone().requires_grad_(True)
+    yb = ops.dense(xb, w, b) + res
+    yb.backward(dy)
+    assert rel_err(xa.grad.float().cpu(), xb.grad.float().cpu()) < 1e-5
+
+
+def test_norm_params_fp32_grads():
+    """GN/RMS take fp32 masters directly on GPU: grads come back fp32 with
+    no bf16 shadow round trip."""
+    from flaxdiff_amd.models.common import GroupNorm, RMSNorm
+    gn = GroupNorm(4, 32).to(_dev())
+    x = torch.randn(2, 8, 8, 32, device=_dev()).bfloat16()
+    y = gn(x, silu=True)
+    assert y.dtype == torch.bfloat16
+    y.float().square().mean().backward()
+    assert gn.weight.grad is not None and gn.weight.grad.dtype == torch.float32
+    rn = RMSNorm(64).to(_dev())
+    x2 = torch.randn(16, 64, device=_dev()).bfloat16()
+    rn(x2).float().square().mean().backward()
+    assert rn.weight.grad.dtype == torch.float32
+
+
 @pytest.mark.parametrize("B,H,W,Ci,Co,k,st", CONV_SHAPES[:6] + CONV_SHAPES[8:12])
 def test_conv2d_backward(B, H, W, Ci, Co, k, st):
     torch.manual_seed(4)
