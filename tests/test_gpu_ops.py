@@ -834,12 +834,12 @@ def test_gemm_nt_matches_reference(M, K, N):
     w = (torch.randn(K, N) * 0.1).bfloat16().cuda()
     bias = torch.randn(N).float().cuda()
     wt = w.t().contiguous()
-    y = ext.gemm_nt(x, wt, bias)
+    y = ext.gemm_nt(x, wt, bias, torch.Tensor())
     ref = x.float() @ w.float() + bias
     rel = (y.float() - ref).abs().max().item() / (ref.abs().max().item() + 1e-9)
     assert rel < 3e-2, rel
     # no-bias path
-    y2 = ext.gemm_nt(x, wt, torch.Tensor())
+    y2 = ext.gemm_nt(x, wt, torch.Tensor(), torch.Tensor())
     ref2 = x.float() @ w.float()
     assert (y2.float() - ref2).abs().max().item() / (ref2.abs().max().item() + 1e-9) < 3e-2
 
