@@ -308,7 +308,7 @@ class Upsample(nn.Module):
         out = ops.nearest_upsample_2x(x)
         out = self.conv(out)
         if residual is not None:
-            out = torch.cat([out, residual], dim=-1)
+            out = ops.cat_channels(out, residual)
         return out
 
 
@@ -326,7 +326,7 @@ class Downsample(nn.Module):
         if residual is not None:
             if residual.shape[1] > out.shape[1]:
                 residual = ops.avg_pool_2x(residual)
-            out = torch.cat([out, residual], dim=-1)
+            out = ops.cat_channels(out, residual)
         return out
 
 
@@ -385,7 +385,7 @@ class ResidualBlock(nn.Module):
             out = self.conv2(out) + residual
 
         if extra_features is not None:
-            out = torch.cat([out, extra_features], dim=-1)
+            out = ops.cat_channels(out, extra_features)
         return out
 
 

@@ -20,6 +20,7 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
+from .. import ops
 from .attention import TransformerBlock
 from .common import (ConvLayer, Downsample, FourierEmbedding, GroupNorm,
                      RMSNorm, ResidualBlock, TimeProjection, Upsample)
@@ -180,7 +181,7 @@ class Unet(nn.Module):
 
         for level in self.up_blocks:
             for res, attn in zip(level["res"], level["attn"]):
-                x = torch.cat([x, downs.pop()], dim=-1)
+                x = ops.cat_channels(x, downs.pop())
                 x = res(x, temb)
                 if not isinstance(attn, nn.Identity):
                     x = attn(x, textcontext)
@@ -188,7 +189,7 @@ class Unet(nn.Module):
                 x = level["up"](x)
 
         x = self.conv_mid(x)
-        x = torch.cat([x, downs.pop()], dim=-1)
+        x = ops.cat_channels(x, downs.pop())
         x = self.final_residual(x, temb)
 
         if isinstance(self.conv_out_norm, GroupNorm):

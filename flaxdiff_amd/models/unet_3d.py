@@ -247,7 +247,7 @@ class UNet3D(nn.Module):
         for level in self.up:
             for rb, tc, ab, tb in zip(level["res"], level["tconv"],
                                       level["attn"], level["tattn"]):
-                h = torch.cat([h, stack.pop()], dim=-1)
+                h = ops.cat_channels(h, stack.pop())
                 h = rb(h, t_vec)
                 h = tc(h, T)
                 if not isinstance(ab, nn.Identity):
@@ -257,7 +257,7 @@ class UNet3D(nn.Module):
             if "up" in level:
                 h = level["up"](h)
 
-        h = torch.cat([h, stack.pop()], dim=-1)
+        h = ops.cat_channels(h, stack.pop())
         h = self.final_res(h, t_vec)
         h = self.final_norm(h, silu=True)
         h = self.conv_out(h)

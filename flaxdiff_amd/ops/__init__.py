@@ -328,6 +328,30 @@ def dense(x: torch.Tensor, w: torch.Tensor, b: Optional[torch.Tensor] = None,
     return y
 
 
+class _Cat2Fn(torch.autograd.Function):
+    """Last-dim 2-tensor concat with a one-pass split backward (the UNet
+    skip concat, reference simple_unet.py:132-160)."""
+
+    @staticmethod
+    def forward(ctx, a, b):
+        ctx.ca = a.shape[-1]
+        ctx.cb = b.shape[-1]
+        return _require_ext().cat2_lastdim(a, b)
+
+    @staticmethod
+    def backward(ctx, dy):
+        da, db = _require_ext().split2_lastdim(dy.contiguous(), ctx.ca, ctx.cb)
+        return da, db
+
+
+def cat_channels(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+    """torch.cat([a, b], -1) with fused fwd/bwd kernels on the GPU path."""
+    if _use_hip(a) and a.dtype == torch.bfloat16 and b.dtype == torch.bfloat16 \
+            and (a.shape[-1] + b.shape[-1]) % 8 == 0:
+        return _Cat2Fn.apply(a.contiguous(), b.contiguous())
+    return torch.cat([a, b], dim=-1)
+
+
 def _bias_grad(dy: torch.Tensor) -> torch.Tensor:
     """Column sum over all but the channel dim (coalesced HIP kernel)."""
     d2 = dy.reshape(-1, dy.shape[-1])

@@ -162,7 +162,12 @@ def test_record_shards_roundtrip(tmp_path):
         assert (got["image"] == recs[i]["image"]).all()
         assert got["score"] == recs[i]["score"]
 
-    # registry + sharded loader (2 ranks see disjoint halves)
+    # registry + sharded loader (2 ranks see disjoint halves).
+    # Disable augmentation: the 10% color jitter can collide two constant
+    # fill values (e.g. 10*1.1 == 11), making `seen` flakily lose a record.
+    import os
+    from flaxdiff_amd.data.sources import AUGMENT_MODE_ENV
+    os.environ[AUGMENT_MODE_ENV] = "none"
     register_record_dataset("test-recs", str(tmp_path), image_size=8)
     seen = set()
     for rank in (0, 1):
@@ -172,6 +177,7 @@ def test_record_shards_roundtrip(tmp_path):
             # images are constant-filled: the fill value identifies the record
             seen.update(int(v) for v in batch["image"][:, 0, 0, 0])
     assert len(seen) == 10
+    os.environ.pop(AUGMENT_MODE_ENV, None)
 
     # converter: any source -> shards
     src = SyntheticImageSource(image_size=8, num_samples=7)

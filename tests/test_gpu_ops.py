@@ -260,6 +260,23 @@ def test_dense_fused_residual_add():
     assert rel_err(xa.grad.float().cpu(), xb.grad.float().cpu()) < 1e-5
 
 
+def test_cat_channels_fwd_bwd():
+    """cat_channels == torch.cat(-1) with exact split backward."""
+    torch.manual_seed(14)
+    for Ca, Cb in ((64, 64), (256, 128), (8, 24)):
+        a = torch.randn(2, 4, 4, Ca).bfloat16().to(_dev())
+        b = torch.randn(2, 4, 4, Cb).bfloat16().to(_dev())
+        y = ops.cat_channels(a, b)
+        assert torch.equal(y, torch.cat([a, b], dim=-1))
+        aa = a.clone().requires_grad_(True)
+        bb = b.clone().requires_grad_(True)
+        yy = ops.cat_channels(aa, bb)
+        dy = torch.randn_like(yy)
+        yy.backward(dy)
+        assert torch.equal(aa.grad, dy[..., :Ca])
+        assert torch.equal(bb.grad, dy[..., Ca:])
+
+
 def test_norm_params_fp32_grads():
     """GN/RMS take fp32 masters directly on GPU: grads come back fp32 with
     no bf16 shadow round trip."""

@@ -10,9 +10,9 @@ x = (torch.randn(B, HW, HW, Ci) * 0.5).bfloat16().cuda()
 w = (torch.randn(3, 3, Ci, Co) * 0.1).bfloat16().cuda()
 bias = torch.randn(Co).float().cuda()
 for _ in range(3):
-    ext.conv2d_fwd(x, w, bias, 1)
+    ext.conv2d_fwd(x, w, bias, 1, torch.Tensor(), torch.Tensor())
 torch.cuda.synchronize()
 for _ in range(10):
-    ext.conv2d_fwd(x, w, bias, 1)
+    ext.conv2d_fwd(x, w, bias, 1, torch.Tensor(), torch.Tensor())
 torch.cuda.synchronize()
 print("done")
