@@ -75,9 +75,7 @@ class GEGLU(nn.Module):
         self.inner = dim * mult
 
     def forward(self, x):
-        h = self.proj(x)
-        hx, hg = h.chunk(2, dim=-1)
-        return hx * F.gelu(hg)
+        return ops.geglu(self.proj(x))
 
 
 class FeedForward(nn.Module):

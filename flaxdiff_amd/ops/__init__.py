@@ -328,6 +328,29 @@ def dense(x: torch.Tensor, w: torch.Tensor, b: Optional[torch.Tensor] = None,
     return y
 
 
+class _GEGLUFn(torch.autograd.Function):
+    """y = h[..., :N] * gelu(h[..., N:]) — one fused pass each way
+    (reference attention.py:207-238)."""
+
+    @staticmethod
+    def forward(ctx, h):
+        ctx.save_for_backward(h)
+        return _require_ext().geglu_fwd(h)
+
+    @staticmethod
+    def backward(ctx, dy):
+        (h,) = ctx.saved_tensors
+        return _require_ext().geglu_bwd(dy.contiguous(), h)
+
+
+def geglu(h: torch.Tensor) -> torch.Tensor:
+    """GEGLU gate: split the last dim in half, y = a * gelu(b)."""
+    if _use_hip(h) and h.dtype == torch.bfloat16 and h.shape[-1] % 16 == 0:
+        return _GEGLUFn.apply(h.contiguous())
+    a, b = h.chunk(2, dim=-1)
+    return a * torch.nn.functional.gelu(b)
+
+
 class _Cat2Fn(torch.autograd.Function):
     """Last-dim 2-tensor concat with a one-pass split backward (the UNet
     skip concat, reference simple_unet.py:132-160)."""

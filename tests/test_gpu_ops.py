@@ -277,6 +277,23 @@ def test_cat_channels_fwd_bwd():
         assert torch.equal(bb.grad, dy[..., Ca:])
 
 
+def test_geglu_fused():
+    """geglu(h) == h[:, :N] * gelu(h[:, N:]) fwd + bwd vs fp32 torch."""
+    torch.manual_seed(15)
+    h0 = torch.randn(64, 256, dtype=torch.float64)
+    href = h0.clone().requires_grad_(True)
+    a, b = href.chunk(2, dim=-1)
+    yref = a * torch.nn.functional.gelu(b)
+    dy = torch.randn_like(yref)
+    yref.backward(dy)
+
+    h = h0.bfloat16().to(_dev()).requires_grad_(True)
+    y = ops.geglu(h)
+    assert rel_err(y.float().cpu(), yref) < 3e-2
+    y.backward(dy.bfloat16().to(_dev()))
+    assert rel_err(h.grad.float().cpu(), href.grad) < 3e-2
+
+
 def test_norm_params_fp32_grads():
     """GN/RMS take fp32 masters directly on GPU: grads come back fp32 with
     no bf16 shadow round trip."""
