@@ -183,7 +183,8 @@ def test_conv2d_fused_residual_add(B, H, W, Ci, Co):
     y0 = ops.conv2d(x, w, bias, stride=1) + res
     y1 = ops.conv2d(x, w, bias, stride=1, add=res)
     assert rel_err(y1.float().cpu(), ref) < 3e-2
-    assert rel_err(y1.float().cpu(), ref) <= rel_err(y0.float().cpu(), ref) * 1.5
+    # (the fused add is never WORSE than unfused by more than rounding noise)
+    assert rel_err(y1.float().cpu(), ref) <= rel_err(y0.float().cpu(), ref) + 8e-3
 
     xa = x.clone().requires_grad_(True)
     wa = w.clone().requires_grad_(True)
