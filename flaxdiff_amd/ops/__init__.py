@@ -673,7 +673,8 @@ def s5_scan(a: torch.Tensor, bu: torch.Tensor) -> torch.Tensor:
 
 def fused_adamw_ema(params_f32, grads, exp_avg, exp_avg_sq, ema_f32, params_bf16,
                     *, lr, beta1, beta2, eps, weight_decay, step, ema_decay,
-                    grad_scale: float = 1.0, scale_dev=None, skip_ctr=None):
+                    grad_scale: float = 1.0, scale_dev=None, skip_ctr=None,
+                    step_dev=None):
     """In-place AdamW + EMA lerp over flat fp32 master buffers.
 
     grads may be bf16 (gets scaled by grad_scale, e.g. 1/world_size folded in).
@@ -687,4 +688,4 @@ def fused_adamw_ema(params_f32, grads, exp_avg, exp_avg_sq, ema_f32, params_bf16
     ext.fused_adamw_ema(params_f32, grads, exp_avg, exp_avg_sq, ema_f32,
                         params_bf16 if params_bf16 is not None else torch.Tensor(),
                         lr, beta1, beta2, eps, weight_decay, step, ema_decay,
-                        grad_scale, scale_dev, skip_ctr)
+                        grad_scale, scale_dev, skip_ctr, step_dev)

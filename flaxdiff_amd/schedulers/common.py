@@ -51,6 +51,16 @@ class NoiseScheduler:
             timesteps = key.uniform((batch_size,), 0.0, float(self.max_timesteps), device=device)
         return timesteps, state
 
+    def sample_timesteps_device(self, batch_size, device) -> torch.Tensor:
+        """Graph-safe timestep draw: same distribution as generate_timesteps
+        but through torch's default (capture-aware) CUDA generator — used by
+        the hipGraph-captured train step, where per-call seeded Generators
+        would bake one fixed draw into the graph."""
+        if isinstance(self.max_timesteps, int) and self.max_timesteps > 1:
+            return torch.randint(0, self.max_timesteps, (batch_size,),
+                                 device=device)
+        return torch.rand(batch_size, device=device) * float(self.max_timesteps)
+
     # -- rates / weights -----------------------------------------------------
     def get_weights(self, steps, shape=(-1, 1, 1, 1)) -> torch.Tensor:
         raise NotImplementedError

@@ -80,6 +80,9 @@ class DiscreteNoiseScheduler(NoiseScheduler):
         timesteps = key.randint((batch_size,), 0, self.max_timesteps, device=device)
         return timesteps, state
 
+    def sample_timesteps_device(self, batch_size, device) -> torch.Tensor:
+        return torch.randint(0, self.max_timesteps, (batch_size,), device=device)
+
     def get_p2_weights(self, k, gamma):
         ac = self._tables["alpha_cumprod"]
         return (k + ac / (1 - ac)) ** (-gamma)

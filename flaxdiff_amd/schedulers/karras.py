@@ -54,6 +54,9 @@ class KarrasVENoiseScheduler(GeneralizedNoiseScheduler):
         timesteps, state = super().generate_timesteps(batch_size, state, device=device)
         return timesteps.float(), state
 
+    def sample_timesteps_device(self, batch_size, device) -> torch.Tensor:
+        return super().sample_timesteps_device(batch_size, device).float()
+
 
 class SimpleExpNoiseScheduler(KarrasVENoiseScheduler):
     """karras.py:52-63 — log-spaced sigma table indexed by integer steps."""
@@ -81,3 +84,6 @@ class EDMNoiseScheduler(KarrasVENoiseScheduler):
         state, key = state.get_random_key()
         timesteps = key.normal((batch_size,), device=device)
         return timesteps, state
+
+    def sample_timesteps_device(self, batch_size, device) -> torch.Tensor:
+        return torch.randn(batch_size, device=device)

@@ -36,6 +36,7 @@ class FlatAdamWEMA:
         self.skip_nonfinite = skip_nonfinite
         self._skipped_host = 0
         self._skip_dev: Optional[torch.Tensor] = None  # GPU-path skip counter
+        self._step_dev: Optional[torch.Tensor] = None  # hipGraph-path counter
         self.step_count = 0
 
         params = [p for p in module.parameters() if p.requires_grad]
@@ -129,11 +130,27 @@ class FlatAdamWEMA:
         return n
 
     @torch.no_grad()
-    def step(self, grad_scale: float = 1.0):
+    def step(self, grad_scale: float = 1.0, use_step_dev: bool = False):
         """grad_scale multiplies gradients before use (e.g. 1/world_size so the
-        all-reduce SUM becomes the reference's pmean)."""
+        all-reduce SUM becomes the reference's pmean).
+
+        use_step_dev (hipGraph capture): keep the Adam step counter in a
+        device int32 so bias correction is recomputed per graph REPLAY
+        instead of baked at capture. The caller owns advancing the host
+        `step_count` mirror per replay (checkpointing still works)."""
         self.step_count += 1
         lr = self.current_lr()
+
+        if use_step_dev:
+            assert self.flat.is_cuda and ops.hip_available(), \
+                "use_step_dev needs the GPU fused-optimizer path"
+            assert self.lr_schedule is None, \
+                "hipGraph step requires a constant lr (schedule is host-side)"
+            if self._step_dev is None:
+                self._step_dev = torch.tensor([self.step_count - 1],
+                                              dtype=torch.int32,
+                                              device=self.flat.device)
+            self._step_dev.add_(1)
 
         if self.flat.is_cuda and ops.hip_available():
             # Clip factor and nonfinite gate are built ON DEVICE so the hot
@@ -160,7 +177,8 @@ class FlatAdamWEMA:
                                 eps=self.eps, weight_decay=self.weight_decay,
                                 step=self.step_count, ema_decay=self.ema_decay,
                                 grad_scale=grad_scale, scale_dev=scale_dev,
-                                skip_ctr=self._skip_dev if self.skip_nonfinite else None)
+                                skip_ctr=self._skip_dev if self.skip_nonfinite else None,
+                                step_dev=self._step_dev if use_step_dev else None)
             self._refresh_t()
             return
 
@@ -232,6 +250,8 @@ class FlatAdamWEMA:
             self.flat_bf16.copy_(self.flat)
         self._refresh_t()
         self.step_count = int(sd["step_count"])
+        if self._step_dev is not None:
+            self._step_dev.fill_(self.step_count)
 
 
 def warmup_cosine_schedule(base_lr: float, warmup_steps: int, total_steps: int,
