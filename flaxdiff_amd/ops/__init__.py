@@ -328,6 +328,35 @@ def dense(x: torch.Tensor, w: torch.Tensor, b: Optional[torch.Tensor] = None,
     return y
 
 
+class _WL2LossFn(torch.autograd.Function):
+    """mean(w[b] * 0.5*(pred-target)^2) in one pass each way (the trainer
+    loss, reference diffusion_trainer.py:203-211)."""
+
+    @staticmethod
+    def forward(ctx, pred, target, wb):
+        ctx.save_for_backward(pred, target, wb)
+        return _require_ext().wl2_loss_fwd(pred, target, wb)
+
+    @staticmethod
+    def backward(ctx, go):
+        pred, target, wb = ctx.saved_tensors
+        dpred = _require_ext().wl2_loss_bwd(pred, target, wb,
+                                            go.reshape(1).float())
+        return dpred, None, None
+
+
+def weighted_l2_loss(pred: torch.Tensor, target: torch.Tensor,
+                     weights: torch.Tensor) -> torch.Tensor:
+    """Fused loss = mean(weights * 0.5*(pred-target)^2); weights broadcast
+    per sample [B,1,1,...]. Falls back to torch ops off the bf16 GPU path."""
+    if _use_hip(pred) and pred.dtype == torch.bfloat16 \
+            and target.dtype == torch.bfloat16:
+        wb = weights.reshape(weights.shape[0]).float().contiguous()
+        return _WL2LossFn.apply(pred.contiguous(), target.contiguous(), wb)
+    return (0.5 * (pred.float() - target.float()) ** 2
+            * weights.to(torch.float32)).mean()
+
+
 class _GEGLUFn(torch.autograd.Function):
     """y = h[..., :N] * gelu(h[..., N:]) — one fused pass each way
     (reference attention.py:207-238)."""

@@ -70,6 +70,16 @@ class DiffusionTrainer(SimpleTrainer):
         self._graph_static: Dict[str, torch.Tensor] = {}
 
     # ------------------------------------------------------------------
+
+    def _weighted_loss(self, pred, target, weights):
+        if self.loss_fn is l2_loss and pred.is_cuda \
+                and pred.dtype == torch.bfloat16 \
+                and target.dtype == torch.bfloat16 \
+                and weights.numel() == pred.shape[0]:
+            return ops.weighted_l2_loss(pred, target, weights)
+        return (self.loss_fn(pred.float(), target.float())
+                * weights.to(torch.float32)).mean()
+
     def _graph_eligible(self, batch) -> bool:
         return (self._graph_ok
                 and self.device.type == "cuda"
@@ -118,7 +128,7 @@ class DiffusionTrainer(SimpleTrainer):
         pred = self.model_output_transform.pred_transform(x_t, pred, rates)
         weights = self.noise_schedule.get_weights(
             timesteps, get_coeff_shapes_tuple(images)).to(self.device)
-        loss = (self.loss_fn(pred.float(), target.float()) * weights).mean()
+        loss = self._weighted_loss(pred, target, weights)
         loss.backward()
         self.optimizer.step(grad_scale=1.0, use_step_dev=True)
         return loss.detach()
@@ -238,8 +248,7 @@ class DiffusionTrainer(SimpleTrainer):
         pred = self.model_output_transform.pred_transform(x_t, pred, rates)
 
         weights = self.noise_schedule.get_weights(timesteps, get_coeff_shapes_tuple(images)).to(dev)
-        nloss = self.loss_fn(pred.float(), target.float())
-        loss = (nloss * weights).mean()
+        loss = self._weighted_loss(pred, target, weights)
 
         loss.backward()
         self.grad_sync.sync()

@@ -297,6 +297,26 @@ def test_geglu_fused():
     assert rel_err(h.grad.float().cpu(), href.grad) < 3e-2
 
 
+def test_weighted_l2_loss_fused():
+    """ops.weighted_l2_loss == mean(w*0.5*(p-t)^2) with matching dpred."""
+    torch.manual_seed(16)
+    B = 4
+    p0 = torch.randn(B, 8, 8, 3, dtype=torch.float64)
+    t0 = torch.randn(B, 8, 8, 3, dtype=torch.float64)
+    w0 = torch.rand(B, dtype=torch.float64) + 0.1
+    pr = p0.clone().requires_grad_(True)
+    lref = (0.5 * (pr - t0) ** 2 * w0[:, None, None, None]).mean()
+    lref.backward()
+
+    p = p0.bfloat16().to(_dev()).requires_grad_(True)
+    t = t0.bfloat16().to(_dev())
+    w = w0.float().to(_dev())
+    loss = ops.weighted_l2_loss(p, t, w)
+    assert abs(loss.item() - lref.item()) < 3e-2 * max(abs(lref.item()), 1.0)
+    loss.backward()
+    assert rel_err(p.grad.float().cpu(), pr.grad) < 3e-2
+
+
 def test_norm_params_fp32_grads():
     """GN/RMS take fp32 masters directly on GPU: grads come back fp32 with
     no bf16 shadow round trip."""
