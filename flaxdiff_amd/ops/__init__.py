@@ -325,6 +325,8 @@ def dense(x: torch.Tensor, w: torch.Tensor, b: Optional[torch.Tensor] = None,
     y = torch.matmul(x, w)
     if b is not None:
         y = y + b.to(y.dtype)
+    if add is not None:
+        y = y + add.to(y.dtype)
     return y
 
 
