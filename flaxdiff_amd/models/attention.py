@@ -116,17 +116,19 @@ class BasicTransformerBlock(nn.Module):
             self.norm2 = RMSNorm(query_dim, eps=norm_epsilon)
             self.norm3 = RMSNorm(query_dim, eps=norm_epsilon)
 
-    def forward(self, hidden_states, context=None):
-        # residual adds fused into each sub-block's final GEMM epilogue
+    def forward(self, hidden_states, context=None, add=None):
+        # residual adds fused into each sub-block's final GEMM epilogue;
+        # `add` rides the LAST sub-block's epilogue (the TransformerBlock
+        # outer residual in the UNet's only_pure_attention default)
         if self.only_pure_attention:
-            return self.attention2(hidden_states, context)
+            return self.attention2(hidden_states, context, add=add)
         if not self.use_cross_only:
             hidden_states = self.attention1(self.norm1(hidden_states),
                                             add=hidden_states)
         hidden_states = self.attention2(self.norm2(hidden_states), context,
                                         add=hidden_states)
         hidden_states = self.ff(self.norm3(hidden_states), add=hidden_states)
-        return hidden_states
+        return hidden_states if add is None else hidden_states + add
 
 
 class TransformerBlock(nn.Module):
@@ -164,11 +166,11 @@ class TransformerBlock(nn.Module):
         projected = self.project_in(x) if self.use_projection else x
         if context is None:
             context = projected
-        projected = self.block(projected, context)
         if self.use_projection:
+            projected = self.block(projected, context)
             if self.only_pure_attention or self.explicitly_add_residual:
                 return self.project_out(projected, add=x)
             return self.project_out(projected)
         if self.only_pure_attention or self.explicitly_add_residual:
-            projected = x + projected
-        return projected
+            return self.block(projected, context, add=x)
+        return self.block(projected, context)
