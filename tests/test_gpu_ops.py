@@ -199,8 +199,10 @@ def test_conv2d_fused_residual_add(B, H, W, Ci, Co):
     yb.backward(dy)
     assert rel_err(xa.grad.float().cpu(), xb.grad.float().cpu()) < 1e-5
     # wgrad's split-M reduce uses fp32 atomics when the split exceeds one
-    # z-chunk: two runs of the SAME wgrad differ by atomic ordering (~1e-5)
-    assert rel_err(wa.grad.float().cpu(), wb.grad.float().cpu()) < 2e-4
+    # z-chunk: two runs of the SAME wgrad differ by atomic ordering (up to
+    # ~1e-4 rel on max-normalized error). 1e-3 still catches wiring bugs
+    # (a wrong dw is O(1) off), immune to the ordering jitter.
+    assert rel_err(wa.grad.float().cpu(), wb.grad.float().cpu()) < 1e-3
 
 
 @pytest.mark.parametrize("B,H,W,Ci,Co,st", [
