@@ -179,6 +179,11 @@ class SimpleTrainer:
             with torch.no_grad():
                 self.optimizer.flat.copy_(self.best_state["flat"].to(self.device))
                 self.optimizer.ema.copy_(self.best_state["ema"].to(self.device))
+                # the forward reads the bf16 shadows, not the fp32 masters —
+                # without this refresh the next steps still run on the
+                # corrupted weights until an optimizer step rewrites them
+                self.optimizer.flat_bf16.copy_(self.optimizer.flat)
+                self.optimizer._refresh_t()
         self.optimizer.exp_avg.zero_()
         self.optimizer.exp_avg_sq.zero_()
         if torch.cuda.is_available():
