@@ -339,3 +339,30 @@ def test_registry_push_best_run_comparison(tmp_path):
     assert path is not None and meta["metrics"]["fid"] == 5.0
     import os as _os
     assert _os.path.isdir(path)
+
+
+def test_watchdog_recovery_refreshes_shadows():
+    """_recover() must restore the bf16 shadows the forward actually reads,
+    not just the fp32 masters."""
+    import torch
+    from flaxdiff_amd.models import Unet
+    from flaxdiff_amd.predictors import EpsilonPredictionTransform
+    from flaxdiff_amd.schedulers import CosineNoiseScheduler
+    from flaxdiff_amd.trainer import DiffusionTrainer
+
+    model = Unet(emb_features=32, feature_depths=(8, 16),
+                 attention_configs=(None, None), num_res_blocks=1,
+                 norm_groups=4, context_dim=16)
+    tr = DiffusionTrainer(model, CosineNoiseScheduler(100),
+                          EpsilonPredictionTransform(), name="wdog",
+                          checkpoint_base_path="/tmp/fdiff_wdog",
+                          distributed=False)
+    tr._snapshot_best(0.5)                       # good snapshot
+    with torch.no_grad():
+        tr.optimizer.flat.fill_(float("nan"))    # corrupt masters
+        tr.optimizer.flat_bf16.fill_(float("nan"))
+    tr._recover()
+    assert torch.isfinite(tr.optimizer.flat).all()
+    assert torch.isfinite(tr.optimizer.flat_bf16.float()).all()
+    assert torch.allclose(tr.optimizer.flat_bf16.float(),
+                          tr.optimizer.flat.bfloat16().float())
