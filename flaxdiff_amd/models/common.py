@@ -305,8 +305,11 @@ class Upsample(nn.Module):
         self.conv = Conv(in_features, features, (3, 3), (1, 1))
 
     def forward(self, x, residual=None):
-        out = ops.nearest_upsample_2x(x)
-        out = self.conv(out)
+        if x.is_cuda and x.dtype == torch.bfloat16:
+            # upsample fused into the conv's halo staging (ops.conv2d_upsample2x)
+            out = ops.conv2d_upsample2x(x, self.conv.weight, self.conv.bias)
+        else:
+            out = self.conv(ops.nearest_upsample_2x(x))
         if residual is not None:
             out = ops.cat_channels(out, residual)
         return out

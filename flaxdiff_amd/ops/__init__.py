@@ -359,6 +359,69 @@ def weighted_l2_loss(pred: torch.Tensor, target: torch.Tensor,
             * weights.to(torch.float32)).mean()
 
 
+class _ConvUp2xFn(torch.autograd.Function):
+    """nearest-2x upsample + 3x3 SAME conv as ONE kernel: the upsample is an
+    address remap inside the conv's halo staging (no materialized big
+    tensor on the forward; reference common.py:203-226 Upsample).
+    Backward rematerializes the upsampled activation once for wgrad and
+    folds the upsample adjoint (2x2 window sum) over the conv dgrad."""
+
+    @staticmethod
+    def forward(ctx, x, w, b):
+        ext = _require_ext()
+        wk = _kernel_view(w)
+        y = ext.conv2d_fwd_up2x(x, wk, b if b is not None else torch.Tensor())
+        if y.numel() == 0:
+            raise _ConvUpIneligible
+        ctx.save_for_backward(x, wk)
+        ctx.has_bias = b is not None
+        ctx.w_dtype = w.dtype
+        ctx.b_dtype = b.dtype if b is not None else None
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, wk = ctx.saved_tensors
+        ext = _require_ext()
+        dy = dy.contiguous()
+        H2, W2 = x.shape[1] * 2, x.shape[2] * 2
+        dx = db = dw = None
+        if ctx.needs_input_grad[0]:
+            dx_up = ext.conv2d_dgrad(dy, wk, 1, H2, W2)
+            dx = ext.upsample2x_bwd(dx_up)
+        if ctx.needs_input_grad[1]:
+            x_up = ext.upsample2x_fwd(x)
+            dw, _ = ext.conv2d_wgrad(dy, x_up, 3, 3, 1)
+            dw = dw if ctx.w_dtype == torch.float32 else dw.to(ctx.w_dtype)
+        if ctx.has_bias:
+            db = _bias_grad(dy)
+            if ctx.b_dtype != torch.float32:
+                db = db.to(ctx.b_dtype)
+        return dx, dw, db
+
+
+class _ConvUpIneligibleError(Exception):
+    pass
+
+
+_ConvUpIneligible = _ConvUpIneligibleError()
+
+
+def conv2d_upsample2x(x: torch.Tensor, w: torch.Tensor,
+                      b: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """conv2d(nearest_upsample_2x(x), w, b) with the upsample fused into the
+    conv's staging on the GPU halo path; falls back to the two-op form."""
+    if _use_hip(x) and x.dtype == torch.bfloat16 and _routable_param(w) \
+            and w.shape[0] == 3 and w.shape[1] == 3:
+        try:
+            return _ConvUp2xFn.apply(
+                x.contiguous(), w.contiguous(),
+                b.contiguous() if b is not None else None)
+        except _ConvUpIneligibleError:
+            pass
+    return conv2d(nearest_upsample_2x(x), w, b, stride=1)
+
+
 class _GEGLUFn(torch.autograd.Function):
     """y = h[..., :N] * gelu(h[..., N:]) — one fused pass each way
     (reference attention.py:207-238)."""

@@ -282,6 +282,29 @@ def test_cat_channels_fwd_bwd():
         assert torch.equal(bb.grad, dy[..., Ca:])
 
 
+def test_conv2d_upsample2x_fused():
+    """Fused nearest-2x + conv == two-op composition, fwd + all grads."""
+    torch.manual_seed(17)
+    for B, HW, Ci, Co in ((2, 8, 128, 128), (2, 32, 64, 64), (1, 4, 512, 256)):
+        x = (torch.randn(B, HW, HW, Ci) * 0.5).bfloat16().to(_dev())
+        w = (torch.randn(3, 3, Ci, Co) / (9 * Ci) ** 0.5).bfloat16().to(_dev())
+        b = (torch.randn(Co) * 0.1).bfloat16().to(_dev())
+
+        xa = x.clone().requires_grad_(True)
+        wa = w.clone().requires_grad_(True)
+        ya = ops.conv2d_upsample2x(xa, wa, b)
+        assert ya.shape == (B, 2 * HW, 2 * HW, Co)
+        xb = x.clone().requires_grad_(True)
+        wb = w.clone().requires_grad_(True)
+        yb = ops.conv2d(ops.nearest_upsample_2x(xb), wb, b, stride=1)
+        assert rel_err(ya.float().cpu(), yb.float().cpu()) < 1e-2
+        dy = torch.randn_like(ya)
+        ya.backward(dy)
+        yb.backward(dy)
+        assert rel_err(xa.grad.float().cpu(), xb.grad.float().cpu()) < 1e-2
+        assert rel_err(wa.grad.float().cpu(), wb.grad.float().cpu()) < 1e-3
+
+
 def test_geglu_fused():
     """geglu(h) == h[:, :N] * gelu(h[:, N:]) fwd + bwd vs fp32 torch."""
     torch.manual_seed(15)
