@@ -371,7 +371,7 @@ class _ConvUp2xFn(torch.autograd.Function):
         ext = _require_ext()
         wk = _kernel_view(w)
         y = ext.conv2d_fwd_up2x(x, wk, b if b is not None else torch.Tensor())
-        if y.numel() == 0:
+        if y is None or y.numel() == 0:   # undefined tensor -> None via pybind
             raise _ConvUpIneligible
         ctx.save_for_backward(x, wk)
         ctx.has_bias = b is not None
