@@ -59,3 +59,14 @@ def test_scheduler_device_sampling_distribution():
     assert k.dtype == torch.float32 and k.min() >= 0 and k.max() <= 1.0
     e = EDMNoiseScheduler(1).sample_timesteps_device(4096, "cpu")
     assert abs(e.mean().item()) < 0.2 and abs(e.std().item() - 1) < 0.2
+
+
+def test_conv2d_upsample2x_cpu_fallback():
+    """conv2d_upsample2x == conv2d(nearest_upsample_2x(x)) off the GPU path
+    (also covers non-64-divisible channels where the fused kernel declines)."""
+    torch.manual_seed(2)
+    x = torch.randn(2, 4, 4, 6)
+    w = torch.randn(3, 3, 6, 10) * 0.2
+    b = torch.randn(10) * 0.1
+    ref = ops.conv2d(ops.nearest_upsample_2x(x), w, b, stride=1)
+    assert torch.allclose(ops.conv2d_upsample2x(x, w, b), ref, atol=1e-6)
