@@ -372,7 +372,7 @@ class _ConvUp2xFn(torch.autograd.Function):
         wk = _kernel_view(w)
         y = ext.conv2d_fwd_up2x(x, wk, b if b is not None else torch.Tensor())
         if y is None or y.numel() == 0:   # undefined tensor -> None via pybind
-            raise _ConvUpIneligible
+            raise _ConvUpIneligibleError
         ctx.save_for_backward(x, wk)
         ctx.has_bias = b is not None
         ctx.w_dtype = w.dtype
@@ -402,9 +402,6 @@ class _ConvUp2xFn(torch.autograd.Function):
 
 class _ConvUpIneligibleError(Exception):
     pass
-
-
-_ConvUpIneligible = _ConvUpIneligibleError()
 
 
 def conv2d_upsample2x(x: torch.Tensor, w: torch.Tensor,
