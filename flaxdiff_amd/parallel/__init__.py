@@ -149,12 +149,18 @@ class GradBucketSynchronizer:
         self._arrived = [0] * len(self.buckets)
         self._launched = [False] * len(self.buckets)
         self._works = []
+        # suspended: hooks become no-ops. The hipGraph-captured train step
+        # replays backward without firing collectives (RCCL must stay out of
+        # the captured graph); the caller then reduces with sync_flat().
+        self.suspended = False
 
         if self.enabled:
             for off, n, p in ranges:
                 p.register_post_accumulate_grad_hook(self._hook)
 
     def _hook(self, p):
+        if self.suspended:
+            return
         bi = self._param_bucket[p]
         self._arrived[bi] += 1
         if self._arrived[bi] >= self.buckets[bi][2] and not self._launched[bi]:
@@ -169,7 +175,7 @@ class GradBucketSynchronizer:
 
     def sync(self):
         """Call after backward: launches any straggler buckets and waits."""
-        if not self.enabled:
+        if not self.enabled or self.suspended:
             return
         for bi in range(len(self.buckets)):
             if not self._launched[bi]:
@@ -179,3 +185,17 @@ class GradBucketSynchronizer:
         self._works.clear()
         self._arrived = [0] * len(self.buckets)
         self._launched = [False] * len(self.buckets)
+
+    def sync_flat(self):
+        """All-reduce the whole flat gradient buffer bucket-by-bucket in one
+        burst (async launches, then wait). Used after a hipGraph replay of
+        backward, where the per-param hooks were suspended — same bucket
+        sizing as the overlapped path so xGMI ring bandwidth is unchanged,
+        only the backward overlap is given up."""
+        if not self.enabled:
+            return
+        works = [dist.all_reduce(self.flat_grad[s0:e0], op=dist.ReduceOp.SUM,
+                                 async_op=True)
+                 for s0, e0, _ in self.buckets]
+        for w in works:
+            w.wait()

@@ -81,9 +81,12 @@ class DiffusionTrainer(SimpleTrainer):
                 * weights.to(torch.float32)).mean()
 
     def _graph_eligible(self, batch) -> bool:
+        # Distributed is graphed too, with RCCL OUTSIDE the graph: the
+        # captured replay covers zero_grad+forward+loss+backward (the ~650
+        # launches whose CPU dispatch would dominate small per-rank batches),
+        # then the gradient all-reduce and fused optimizer run eagerly.
         return (self._graph_ok
                 and self.device.type == "cuda"
-                and not self.dist.is_distributed
                 and self.autoencoder is None
                 and self.optimizer.lr_schedule is None
                 and "cond_embs" not in batch
@@ -130,7 +133,9 @@ class DiffusionTrainer(SimpleTrainer):
             timesteps, get_coeff_shapes_tuple(images)).to(self.device)
         loss = self._weighted_loss(pred, target, weights)
         loss.backward()
-        self.optimizer.step(grad_scale=1.0, use_step_dev=True)
+        if not self.dist.is_distributed:
+            # single GPU: the fused optimizer rides inside the graph
+            self.optimizer.step(grad_scale=1.0, use_step_dev=True)
         return loss.detach()
 
     def _train_step_graphed(self, batch) -> Optional[Dict[str, float]]:
@@ -145,6 +150,14 @@ class DiffusionTrainer(SimpleTrainer):
         st = self._graph_static
         if self._graph is None:
             try:
+                if self.dist.is_distributed:
+                    # collectives must not be captured: suspend the bucket
+                    # hooks (sync_flat reduces after each replay instead) and
+                    # decorrelate each rank's capture-aware RNG stream
+                    self.grad_sync.suspended = True
+                    torch.cuda.manual_seed(
+                        (self.rngs.seed * 0x9E3779B1 + self.dist.rank * 7919)
+                        & 0x7FFFFFFF)
                 st["image"] = torch.empty_like(images, device=self.device)
                 st["image"].copy_(images.to(self.device))
                 if torch.is_tensor(text):
@@ -158,26 +171,35 @@ class DiffusionTrainer(SimpleTrainer):
                 g = torch.cuda.CUDAGraph()
                 with torch.cuda.graph(g):
                     st["loss"] = self._graph_core()
-                # the capture pass runs the python (step_count += 1) but
-                # records, not executes, the _step_dev add — undo the mirror
-                self.optimizer.step_count -= 1
+                # replays never run python autograd hooks, so the bucket
+                # hooks only needed suspending for warmup+capture — restore
+                # them now (an eager-fallback step later must reduce!)
+                self.grad_sync.suspended = False
+                if not self.dist.is_distributed:
+                    # the ws=1 capture pass runs the python (step_count += 1)
+                    # but records, not executes, the _step_dev add — undo
+                    self.optimizer.step_count -= 1
                 # RNG freshness self-check: replays must draw NEW noise via
                 # the capture-aware generator; identical losses mean frozen
                 # RNG (unsupported) — fall back to eager permanently.
                 g.replay()
                 l1 = float(st["loss"])
+                self._post_replay()
                 g.replay()
                 l2 = float(st["loss"])
-                self.optimizer.step_count += 2
+                self._post_replay()
+                if not self.dist.is_distributed:
+                    self.optimizer.step_count += 2   # _step_dev ran in-graph
                 if l1 == l2:
                     raise RuntimeError("graph RNG frozen")
                 self._graph = g
                 self._graph_key = key
-                return {"loss": l2}
+                return {"loss": self._graph_loss()}
             except Exception:
                 self._graph_ok = False       # capture unsupported: stay eager
                 self._graph = None
                 self._graph_static = {}
+                self.grad_sync.suspended = False
                 return None
         else:
             st["image"].copy_(images.to(self.device, non_blocking=True))
@@ -185,8 +207,24 @@ class DiffusionTrainer(SimpleTrainer):
                 st["text"].copy_(text.to(self.device, self.compute_dtype,
                                          non_blocking=True))
         self._graph.replay()
-        self.optimizer.step_count += 1       # host mirror of _step_dev
-        return {"loss": float(st["loss"])}
+        self._post_replay()
+        if not self.dist.is_distributed:
+            self.optimizer.step_count += 1   # host mirror of _step_dev
+        return {"loss": self._graph_loss()}
+
+    def _post_replay(self):
+        """Distributed epilogue of a graph replay: reduce the flat gradient
+        buffer (same buckets as the overlapped eager path) and run the fused
+        optimizer eagerly. Single-GPU replays carry the optimizer in-graph."""
+        if self.dist.is_distributed:
+            self.grad_sync.sync_flat()
+            self.optimizer.step(grad_scale=1.0 / self.dist.world_size)
+
+    def _graph_loss(self) -> float:
+        loss = self._graph_static["loss"].detach()
+        if self.dist.is_distributed:
+            loss = parallel.all_reduce_mean_scalar(loss.clone())
+        return float(loss)
 
     # ------------------------------------------------------------------
     def train_step(self, batch: Dict[str, torch.Tensor]) -> Dict[str, float]:

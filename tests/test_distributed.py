@@ -285,3 +285,81 @@ def test_bench_dry_run_torchrun_cpu():
     assert rec["n_gpus"] == 4
     assert rec["steps"] == 2
     assert rec["value"] > 0
+
+
+def _syncflat_worker(rank, world, port, results):
+    os.environ.update({
+        "RANK": str(rank), "WORLD_SIZE": str(world), "LOCAL_RANK": str(rank),
+        "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+    })
+    import torch.distributed as dist
+    from flaxdiff_amd.predictors import EpsilonPredictionTransform
+    from flaxdiff_amd.schedulers import CosineNoiseScheduler
+    from flaxdiff_amd.trainer import DiffusionTrainer
+
+    torch.manual_seed(7)
+    m = Unet(emb_features=32, feature_depths=[8, 16],
+             attention_configs=[None, None], num_res_blocks=1,
+             num_middle_res_blocks=1, norm_groups=4, context_dim=16)
+    tr = DiffusionTrainer(m, CosineNoiseScheduler(1000),
+                          EpsilonPredictionTransform(), name="syncflat",
+                          checkpoint_base_path="/tmp/fdiff_syncflat",
+                          text_context_shape=(4, 16), distributed=True)
+    g = torch.Generator().manual_seed(300 + rank)
+    batch = {"image": torch.randint(0, 255, (4, 16, 16, 3), generator=g,
+                                    dtype=torch.uint8)}
+    # the hipGraph-replay protocol: hooks suspended so backward fires no
+    # collectives; sync_flat() reduces the whole flat buffer afterwards
+    tr.grad_sync.suspended = True
+    tr.optimizer.zero_grad()
+    images = batch["image"].float()
+    images = (images - 127.5) / 127.5
+    B = images.shape[0]
+    text = tr.null_context.unsqueeze(0).expand(B, *tr.null_context.shape)
+    from flaxdiff_amd.utils import get_coeff_shapes_tuple
+    timesteps = tr.noise_schedule.sample_timesteps_device(B, "cpu")
+    noise = torch.randn(images.shape)
+    rates = tr.noise_schedule.get_rates(timesteps,
+                                        get_coeff_shapes_tuple(images))
+    x_t, c_in, target = tr.model_output_transform.forward_diffusion(
+        images, noise, rates)
+    x_in, t_in = tr.noise_schedule.transform_inputs(x_t * c_in, timesteps)
+    pred = tr.model(x_in, t_in, text)
+    pred = tr.model_output_transform.pred_transform(x_t, pred, rates)
+    w = tr.noise_schedule.get_weights(timesteps,
+                                      get_coeff_shapes_tuple(images))
+    ((0.5 * (pred.float() - target.float()) ** 2) * w).mean().backward()
+    # no collective fired yet: local grads differ across ranks
+    local = tr.optimizer.flat_grad.clone()
+    tr.grad_sync.sync_flat()
+    tr.grad_sync.suspended = False
+    reduced = tr.optimizer.flat_grad.clone()
+    gathered = [torch.zeros_like(reduced) for _ in range(world)]
+    dist.all_gather(gathered, reduced)
+    same = all(torch.allclose(gathered[0], gi, atol=1e-6) for gi in gathered)
+    results[rank] = {"reduced_equal": bool(same),
+                     "was_local": not torch.allclose(local, reduced)}
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_sync_flat_matches_reduction():
+    """sync_flat() (the graph-replay reduction path) produces identical
+    summed grads on every rank, and suspended hooks fire no collectives."""
+    world = 2
+    ctx = mp.get_context("spawn")
+    with ctx.Manager() as man:
+        results = man.dict()
+        port = 29741
+        procs = [ctx.Process(target=_syncflat_worker,
+                             args=(r, world, port, results))
+                 for r in range(world)]
+        for p in procs:
+            p.start()
+        for p in procs:
+            p.join(timeout=240)
+        for p in procs:
+            assert p.exitcode == 0, f"worker failed: {p.exitcode}"
+        res = dict(results)
+    assert res[0]["reduced_equal"] and res[1]["reduced_equal"]
+    assert res[0]["was_local"] and res[1]["was_local"]
