@@ -149,6 +149,8 @@ class DiffusionTrainer(SimpleTrainer):
             return None                      # shape changed: eager fallback
         st = self._graph_static
         if self._graph is None:
+            # --- capture attempt: NO collectives may run inside ----------
+            g = None
             try:
                 if self.dist.is_distributed:
                     # collectives must not be captured: suspend the bucket
@@ -161,7 +163,8 @@ class DiffusionTrainer(SimpleTrainer):
                 st["image"] = torch.empty_like(images, device=self.device)
                 st["image"].copy_(images.to(self.device))
                 if torch.is_tensor(text):
-                    st["text"] = text.to(self.device, self.compute_dtype)                         .clone()
+                    st["text"] = text.to(self.device,
+                                         self.compute_dtype).clone()
                 side = torch.cuda.Stream()
                 side.wait_stream(torch.cuda.current_stream())
                 with torch.cuda.stream(side):
@@ -171,36 +174,43 @@ class DiffusionTrainer(SimpleTrainer):
                 g = torch.cuda.CUDAGraph()
                 with torch.cuda.graph(g):
                     st["loss"] = self._graph_core()
-                # replays never run python autograd hooks, so the bucket
-                # hooks only needed suspending for warmup+capture — restore
-                # them now (an eager-fallback step later must reduce!)
-                self.grad_sync.suspended = False
                 if not self.dist.is_distributed:
                     # the ws=1 capture pass runs the python (step_count += 1)
                     # but records, not executes, the _step_dev add — undo
                     self.optimizer.step_count -= 1
-                # RNG freshness self-check: replays must draw NEW noise via
-                # the capture-aware generator; identical losses mean frozen
-                # RNG (unsupported) — fall back to eager permanently.
-                g.replay()
-                l1 = float(st["loss"])
-                self._post_replay()
-                g.replay()
-                l2 = float(st["loss"])
-                self._post_replay()
-                if not self.dist.is_distributed:
-                    self.optimizer.step_count += 2   # _step_dev ran in-graph
-                if l1 == l2:
-                    raise RuntimeError("graph RNG frozen")
-                self._graph = g
-                self._graph_key = key
-                return {"loss": self._graph_loss()}
             except Exception:
-                self._graph_ok = False       # capture unsupported: stay eager
+                g = None
+            # replays never run python autograd hooks, so the bucket hooks
+            # only needed suspending for warmup+capture — restore them (an
+            # eager-fallback step must overlap its reduction again)
+            self.grad_sync.suspended = False
+            # --- all-ranks agreement: a rank-split graph/eager mix would
+            # interleave mismatched collectives and deadlock -------------
+            if not self._all_ranks_agree(g is not None):
+                self._graph_ok = False
                 self._graph = None
                 self._graph_static = {}
-                self.grad_sync.suspended = False
                 return None
+            # RNG freshness self-check: replays must draw NEW noise via the
+            # capture-aware generator; identical losses mean frozen RNG.
+            # The verdict is software-global (same build on every rank), so
+            # ranks stay in lockstep through the agreement below.
+            g.replay()
+            l1 = float(st["loss"])
+            self._post_replay()
+            g.replay()
+            l2 = float(st["loss"])
+            self._post_replay()
+            if not self.dist.is_distributed:
+                self.optimizer.step_count += 2       # _step_dev ran in-graph
+            if not self._all_ranks_agree(l1 != l2):
+                self._graph_ok = False
+                self._graph = None
+                self._graph_static = {}
+                return None
+            self._graph = g
+            self._graph_key = key
+            return {"loss": self._graph_loss()}
         else:
             st["image"].copy_(images.to(self.device, non_blocking=True))
             if "text" in st and torch.is_tensor(text):
@@ -211,6 +221,14 @@ class DiffusionTrainer(SimpleTrainer):
         if not self.dist.is_distributed:
             self.optimizer.step_count += 1   # host mirror of _step_dev
         return {"loss": self._graph_loss()}
+
+    def _all_ranks_agree(self, ok: bool) -> bool:
+        if not self.dist.is_distributed:
+            return ok
+        import torch.distributed as dist
+        flag = torch.tensor([1.0 if ok else 0.0], device=self.device)
+        dist.all_reduce(flag, op=dist.ReduceOp.MIN)
+        return bool(flag.item() >= 1.0)
 
     def _post_replay(self):
         """Distributed epilogue of a graph replay: reduce the flat gradient
