@@ -214,3 +214,21 @@ def test_voxceleb2_like_av_source(tmp_path):
     assert (s["instance_masked_images"][:, 8:] == 0).all()
     assert s["identity"] == "id001"
     assert s["mels"].shape[0] == 20 and s["mels"].shape[1] > 10
+
+
+def test_benchmark_loader_cli():
+    """The loader throughput/leak benchmark CLI runs end to end (reference
+    benchmark_decord.py / training.py --dataset_test equivalents)."""
+    import json
+    import io
+    import contextlib
+    from flaxdiff_amd.data.benchmark_loader import main
+
+    buf = io.StringIO()
+    with contextlib.redirect_stdout(buf):
+        main(["--dataset", "synthetic-64", "--batches", "4",
+              "--batch_size", "4", "--workers", "0", "--report_every", "2"])
+    lines = [l for l in buf.getvalue().splitlines() if l.startswith("{")]
+    assert lines, buf.getvalue()
+    rec = json.loads(lines[-1])
+    assert rec["images_per_sec"] > 0 and "rss_drift_mb" in rec
