@@ -366,3 +366,33 @@ def test_watchdog_recovery_refreshes_shadows():
     assert torch.isfinite(tr.optimizer.flat_bf16.float()).all()
     assert torch.allclose(tr.optimizer.flat_bf16.float(),
                           tr.optimizer.flat.bfloat16().float())
+
+
+def test_graph_eligibility_predicate():
+    """_graph_eligible gates: CPU device, autoencoder, lr schedules and
+    multi-modality batches must all force the eager path."""
+    import torch
+    from flaxdiff_amd.models import Unet
+    from flaxdiff_amd.predictors import EpsilonPredictionTransform
+    from flaxdiff_amd.schedulers import CosineNoiseScheduler
+    from flaxdiff_amd.trainer import DiffusionTrainer
+
+    model = Unet(emb_features=32, feature_depths=(8, 16),
+                 attention_configs=(None, None), num_res_blocks=1,
+                 norm_groups=4, context_dim=16)
+    tr = DiffusionTrainer(model, CosineNoiseScheduler(100),
+                          EpsilonPredictionTransform(), name="gelig",
+                          checkpoint_base_path="/tmp/fdiff_gelig",
+                          distributed=False)
+    img = torch.randint(0, 255, (2, 16, 16, 3), dtype=torch.uint8)
+    # CPU device -> ineligible (hipGraphs are a GPU feature)
+    assert not tr._graph_eligible({"image": img})
+    # multi-modality conditioning -> ineligible
+    assert not tr._graph_eligible({"image": img, "cond_embs": [img]})
+    # lr schedule is host-side state -> ineligible
+    tr.optimizer.lr_schedule = lambda s: 1e-4
+    assert not tr._graph_eligible({"image": img})
+    tr.optimizer.lr_schedule = None
+    # env kill switch
+    tr._graph_ok = False
+    assert not tr._graph_eligible({"image": img})
